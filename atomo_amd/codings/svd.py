@@ -1,0 +1,253 @@
+"""ATOMO's atomic sparsification codec: SVD + unbiased importance sampling.
+
+Semantics (what, not how) follow /root/reference/src/codings/svd.py:
+  * ``grad_to_2d``       — svd.py:12-28  (gradient reshaped to a 2-D matrix)
+  * ``sample_svd``       — svd.py:49-67  (Bernoulli draws, p_i = min(1, r*s_i/sum(s)),
+                           kept atoms rescaled s_i/p_i for unbiasedness; sum(p) = r
+                           is the paper's sparsity budget)
+  * encode/decode        — svd.py:79-118 / 160-178
+
+MI355X-first differences from the reference:
+  * everything stays on-device in fp32; factors are written into a fixed
+    wire region ``[r_hat | u^T (r_max x m) | s (r_max) | v^T (r_max x n)]``
+    so a whole model's packets ship as one RCCL gather (no pickle).
+  * decode is a fused rank-k accumulate (``out2d += (u*s) @ vT``) straight
+    into the PS aggregation buffer — hipBLASLt GEMM or the HIP kernel in
+    ops/csrc/svd_kernels.hip, never a host hop.
+  * factorization backend is pluggable: ``torch`` (torch.linalg.svd, the
+    oracle) or ``gram`` (one-pass Gram + Jacobi eigensolver on MFMA — the
+    tall-skinny shapes here have n <= ~64, so A^T A is tiny and exact).
+  * odd-length 1-D gradients are zero-padded to even (the reference's
+    ``n//2`` reshape crashes on odd sizes — fixed with intent).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, Optional, Sequence, Tuple
+
+import torch
+
+from .base import Codec, LayerSpec
+
+
+def _fold2d_shape(shape: Sequence[int]) -> Tuple[int, int, int]:
+    """Return (m, n, padded_numel) of the 2-D fold of ``shape``.
+
+    1-D (k,)        -> (ceil(k/2), 2)            [zero-pad if k odd]
+    2-D (a, b)      -> (a, b)
+    nD  (a, b, *c)  -> (a*b/2, 2*prod(c)) when a*b even else (a*b, prod(c));
+                       trailing singleton dims collapse to (a, b).
+    """
+    shape = tuple(int(s) for s in shape)
+    if len(shape) == 1:
+        k = shape[0]
+        m = (k + 1) // 2
+        return m, 2, m * 2
+    if len(shape) == 2:
+        return shape[0], shape[1], shape[0] * shape[1]
+    rest = 1
+    for s in shape[2:]:
+        rest *= s
+    a, b = shape[0], shape[1]
+    if rest == 1:
+        return a, b, a * b
+    ab = a * b
+    if ab % 2 == 0:
+        return ab // 2, 2 * rest, ab * rest
+    return ab, rest, ab * rest
+
+
+def grad_to_2d(grad: torch.Tensor) -> torch.Tensor:
+    """Reshape a gradient to the 2-D matrix the SVD operates on."""
+    m, n, padded = _fold2d_shape(grad.shape)
+    flat = grad.reshape(-1)
+    if padded != flat.numel():
+        flat = torch.cat([flat, flat.new_zeros(padded - flat.numel())])
+    return flat.view(m, n)
+
+
+def sample_svd(
+    s: torch.Tensor,
+    rank: int = 0,
+    generator: Optional[torch.Generator] = None,
+    max_atoms: Optional[int] = None,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Importance-sample singular-value indices.
+
+    p_i = min(1, rank * s_i / sum(s)) (or s_i/s_0 when rank == 0); one
+    Bernoulli draw per singular value; redraw until non-empty.  Returns
+    (sampled_idx int64, probs fp32) on the CPU (the vectors are tiny —
+    n_sv <= ~64).  Unbiasedness invariant: E[sum_{i in S} (s_i/p_i) u_i v_i^T]
+    equals the exact gradient.
+    """
+    s = s.detach().to("cpu", torch.float32)
+    if s.numel() == 0:
+        return torch.zeros(0, dtype=torch.int64), torch.zeros(0)
+    if float(s[0]) < 1e-6:
+        return torch.tensor([0]), torch.tensor([1.0])
+    if rank == 0:
+        probs = s / s[0]
+    else:
+        probs = rank * s / s.sum()
+    probs = probs.clamp(max=1.0)
+    for _ in range(64):
+        draws = torch.rand(probs.shape, generator=generator) < probs
+        idx = draws.nonzero(as_tuple=False).flatten()
+        if idx.numel() > 0:
+            break
+    else:  # pragma: no cover — p_0 == 1 makes this unreachable in practice
+        idx = torch.tensor([0])
+    if max_atoms is not None and idx.numel() > max_atoms:
+        # keep the highest-probability atoms (s is sorted descending, so
+        # these are the first ones); overflow of the fixed wire budget is a
+        # tail event counted by the runtime.
+        idx = idx[:max_atoms]
+    return idx, probs[idx]
+
+
+class SVDCodec(Codec):
+    name = "svd"
+
+    def __init__(
+        self,
+        rank: int = 3,
+        random_sample: bool = True,
+        compress: bool = True,
+        backend: str = "auto",
+        max_atoms: Optional[int] = None,
+        generator: Optional[torch.Generator] = None,
+        **kwargs,
+    ):
+        self.rank = int(rank)
+        self.random_sample = bool(random_sample)
+        self.compress = bool(compress)
+        self.backend = backend
+        self.generator = generator
+        # fixed wire budget per layer: sampled-atom count is Poisson-binomial
+        # with mean <= rank, so mean + generous tail covers ~all steps.
+        self._max_atoms = max_atoms
+        self.overflow_count = 0  # times the budget truncated a sample
+
+    # -- layout ----------------------------------------------------------
+    def _r_max(self, n_sv: int) -> int:
+        if self._max_atoms is not None:
+            return min(n_sv, self._max_atoms)
+        if self.rank == 0:
+            return n_sv
+        return min(n_sv, max(8, 3 * self.rank + 4))
+
+    def layer_meta(self, shape: Sequence[int]) -> dict:
+        m, n, padded = _fold2d_shape(shape)
+        n_sv = min(m, n)
+        return {"m": m, "n": n, "padded": padded, "r_max": self._r_max(n_sv)}
+
+    def wire_words(self, shape: Sequence[int]) -> int:
+        if not self.compress:
+            return int(torch.Size(shape).numel())
+        meta = self.layer_meta(shape)
+        return 1 + meta["r_max"] * (meta["m"] + meta["n"] + 1)
+
+    # -- factorization ---------------------------------------------------
+    def factorize(self, a2d: torch.Tensor):
+        """Thin SVD of the reshaped gradient.  Returns (U, S, Vh)."""
+        backend = self.backend
+        if backend == "auto":
+            backend = "gram" if a2d.is_cuda else "torch"
+        if backend == "gram" and a2d.is_cuda:
+            from ..ops import svd_ops
+
+            return svd_ops.gram_svd(a2d)
+        return torch.linalg.svd(a2d, full_matrices=False)
+
+    # -- oracle API ------------------------------------------------------
+    def encode(self, grad: torch.Tensor) -> Dict:
+        if not self.compress:
+            return {"grad": grad, "encode": False}
+        orig_size = list(grad.shape)
+        a2d = grad_to_2d(grad.float())
+        u, s, vh = self.factorize(a2d)
+        if self.random_sample:
+            idx, probs = sample_svd(s, rank=self.rank, generator=self.generator)
+            idx_dev = idx.to(u.device)
+            u = u[:, idx_dev]
+            s = s[idx_dev] / probs.to(s.device)
+            vh = vh[idx_dev, :]
+        elif self.rank > 0:
+            u = u[:, : self.rank]
+            s = s[: self.rank]
+            vh = vh[: self.rank, :]
+        return {
+            "u": u,
+            "s": s,
+            "vT": vh,
+            "orig_size": orig_size,
+            "encode": True,
+            "rank": self.rank,
+        }
+
+    def decode(self, code: Dict) -> torch.Tensor:
+        if not code.get("encode", False):
+            return code["grad"]
+        u, s, vh = code["u"], code["s"], code["vT"]
+        grad2d = (u * s.unsqueeze(0)) @ vh
+        numel = 1
+        for d in code["orig_size"]:
+            numel *= d
+        return grad2d.reshape(-1)[:numel].view(code["orig_size"])
+
+    # -- wire API --------------------------------------------------------
+    def encode_into(self, grad: torch.Tensor, region: torch.Tensor, spec: LayerSpec) -> int:
+        if not self.compress:
+            region[: spec.numel].copy_(grad.reshape(-1))
+            return spec.numel
+        meta = spec.meta
+        m, n, r_max = meta["m"], meta["n"], meta["r_max"]
+        a2d = grad_to_2d(grad.float())
+        u, s, vh = self.factorize(a2d)
+        if self.random_sample:
+            idx, probs = sample_svd(
+                s, rank=self.rank, generator=self.generator, max_atoms=None
+            )
+            if idx.numel() > r_max:
+                self.overflow_count += 1
+                idx, probs = idx[:r_max], probs[:r_max]
+            idx_dev = idx.to(u.device)
+            u = u[:, idx_dev]
+            s = s[idx_dev] / probs.to(s.device)
+            vh = vh[idx_dev, :]
+        else:
+            r = min(self.rank, r_max) if self.rank > 0 else r_max
+            u, s, vh = u[:, :r], s[:r], vh[:r, :]
+        r_hat = s.numel()
+        region[0] = float(r_hat)
+        off = 1
+        region[off : off + r_hat * m].copy_(u.t().reshape(-1))  # u^T row-major
+        off += r_max * m
+        region[off : off + r_hat].copy_(s)
+        off += r_max
+        region[off : off + r_hat * n].copy_(vh.reshape(-1))
+        return 1 + r_hat * (m + n + 1)
+
+    def decode_from(self, region: torch.Tensor, out: torch.Tensor, spec: LayerSpec) -> None:
+        if not self.compress:
+            out += region[: spec.numel]
+            return
+        meta = spec.meta
+        m, n, r_max = meta["m"], meta["n"], meta["r_max"]
+        r_hat = int(region[0].item())
+        if r_hat == 0:
+            return
+        off = 1
+        ut = region[off : off + r_hat * m].view(r_hat, m)
+        off = 1 + r_max * m
+        s = region[off : off + r_hat]
+        off = 1 + r_max * m + r_max
+        vh = region[off : off + r_hat * n].view(r_hat, n)
+        if spec.numel == meta["padded"]:
+            out2d = out.view(m, n)
+            # fused rank-k accumulate: out2d += u @ diag(s) @ vT
+            out2d.addmm_((ut.t() * s.unsqueeze(0)), vh)
+        else:
+            grad2d = (ut.t() * s.unsqueeze(0)) @ vh
+            out += grad2d.reshape(-1)[: spec.numel]

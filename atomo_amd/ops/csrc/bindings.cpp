@@ -1,0 +1,112 @@
+// Python bindings for the atomo_amd gfx950 kernels (atomo_kernels.hip).
+// Compiled as C++ with the torch extension glue; kernel launches are the
+// extern "C" functions in the .hip translation unit.
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime_api.h>
+
+#include <cstdint>
+
+extern "C" {
+void atomo_qsgd_pack_launch(const float*, float*, uint32_t*, int64_t, int, int,
+                            bool, uint64_t, int, int, hipStream_t);
+void atomo_qsgd_unpack_acc_launch(const float*, const uint32_t*, float*,
+                                  int64_t, int, int, int, int, hipStream_t);
+void atomo_svd_decode_acc_launch(const float*, float*, int, int64_t, int, int,
+                                 int, hipStream_t);
+void atomo_fused_sgd_launch(float*, const float*, float*, int64_t, float,
+                            float, float, bool, float, float, hipStream_t);
+}
+
+namespace {
+
+void check_f32_cuda(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be a CUDA (ROCm) tensor");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void qsgd_pack(torch::Tensor grad, torch::Tensor region, int64_t bucket_size,
+               int64_t qlevel, bool terngrad, int64_t seed) {
+  check_f32_cuda(grad, "grad");
+  check_f32_cuda(region, "region");
+  const int64_t numel = grad.numel();
+  const int bits = 1 + (int)qlevel;
+  const int epw = 32 / bits;
+  const int nb = (int)((numel + bucket_size - 1) / bucket_size);
+  const int wpb = (int)((bucket_size + epw - 1) / epw);
+  TORCH_CHECK(region.numel() >= nb + (int64_t)nb * wpb, "region too small");
+  TORCH_CHECK(bucket_size <= 8192, "bucket_size too large for LDS staging");
+  float* norms = region.data_ptr<float>();
+  uint32_t* packed = reinterpret_cast<uint32_t*>(norms + nb);
+  atomo_qsgd_pack_launch(grad.data_ptr<float>(), norms, packed, numel,
+                         (int)bucket_size, (int)qlevel, terngrad,
+                         (uint64_t)seed, nb, wpb, cur_stream());
+}
+
+void qsgd_unpack_acc(torch::Tensor region, torch::Tensor out, int64_t numel,
+                     int64_t bucket_size, int64_t qlevel) {
+  check_f32_cuda(region, "region");
+  check_f32_cuda(out, "out");
+  const int bits = 1 + (int)qlevel;
+  const int epw = 32 / bits;
+  const int nb = (int)((numel + bucket_size - 1) / bucket_size);
+  const int wpb = (int)((bucket_size + epw - 1) / epw);
+  TORCH_CHECK(out.numel() >= numel, "out too small");
+  TORCH_CHECK(region.numel() >= nb + (int64_t)nb * wpb, "region too small");
+  const float* norms = region.data_ptr<float>();
+  const uint32_t* packed = reinterpret_cast<const uint32_t*>(norms + nb);
+  atomo_qsgd_unpack_acc_launch(norms, packed, out.data_ptr<float>(), numel,
+                               (int)bucket_size, (int)qlevel, nb, wpb,
+                               cur_stream());
+}
+
+void svd_decode_acc(torch::Tensor regions, torch::Tensor out2d, int64_t m,
+                    int64_t n, int64_t r_max) {
+  // regions may be a narrow() view of the stacked (W, total_wire) gather
+  // buffer: row-contiguous with an arbitrary row stride.
+  TORCH_CHECK(regions.is_cuda(), "regions must be a CUDA (ROCm) tensor");
+  TORCH_CHECK(regions.scalar_type() == torch::kFloat32, "regions must be fp32");
+  check_f32_cuda(out2d, "out2d");
+  TORCH_CHECK(regions.dim() == 2, "regions must be (W, wire_words)");
+  TORCH_CHECK(regions.stride(1) == 1, "regions rows must be contiguous");
+  const int W = (int)regions.size(0);
+  const int64_t stride = regions.stride(0);
+  TORCH_CHECK(regions.size(1) >= 1 + r_max * (m + n + 1), "region too small");
+  TORCH_CHECK(out2d.numel() == m * n, "out2d shape mismatch");
+  atomo_svd_decode_acc_launch(regions.data_ptr<float>(),
+                              out2d.data_ptr<float>(), W, stride, (int)m,
+                              (int)n, (int)r_max, cur_stream());
+}
+
+void fused_sgd(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
+               double momentum, double weight_decay, bool nesterov,
+               double dampening, double grad_scale) {
+  check_f32_cuda(p, "p");
+  check_f32_cuda(g, "g");
+  const int64_t n = p.numel();
+  TORCH_CHECK(g.numel() == n, "grad size mismatch");
+  if (momentum != 0.0)
+    TORCH_CHECK(buf.numel() == n, "momentum buffer size mismatch");
+  atomo_fused_sgd_launch(p.data_ptr<float>(), g.data_ptr<float>(),
+                         buf.numel() ? buf.data_ptr<float>() : nullptr, n,
+                         (float)lr, (float)momentum, (float)weight_decay,
+                         nesterov, (float)dampening, (float)grad_scale,
+                         cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("qsgd_pack", &qsgd_pack, "QSGD bucket quantize+pack (gfx950)");
+  m.def("qsgd_unpack_acc", &qsgd_unpack_acc, "QSGD unpack+accumulate");
+  m.def("svd_decode_acc", &svd_decode_acc,
+        "fused rank-k SVD decode+accumulate over W packets");
+  m.def("fused_sgd", &fused_sgd, "fused flat SGD apply");
+}

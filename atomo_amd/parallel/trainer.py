@@ -1,0 +1,242 @@
+"""Synchronous parameter-server trainer.
+
+One object runs on every rank; role by rank (reference step machines:
+SyncReplicasMaster_NN.train, sync_replicas_master_nn.py:173-234, and
+DistributedWorker.train, distributed_worker.py:166-262).  Per global step:
+
+  1. fetch   broadcast of the flat fp32 parameter buffer from rank 0
+             (replaces per-layer fp64 Bcast + step handshake tags 10).
+  2. comp    forward/backward on this rank's synthetic batch (skipped by a
+             dedicated PS).
+  3. encode  per-layer codec encode straight into the fixed wire bucket.
+  4. comm    ONE gather of wire buckets to rank 0 (or ONE reduce for the
+             raw codec — RCCL sums on the wire).
+  5. decode  PS accumulates all workers' packets into the flat agg buffer
+             (fused HIP kernels on GPU).
+  6. apply   external-grad optimizer step with grad_scale = 1/num_workers,
+             then lr shrinkage every ``shrink_freq`` steps (reference
+             hardcodes 50, sync_replicas_master_nn.py:103-107).
+
+Full-sync semantics are preserved: every rank participates in both
+collectives every step, so the PS implicitly collects exactly
+``num_workers`` gradients per step (reference counter semantics,
+sync_replicas_master_nn.py:113,212-214).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from ..codings import Codec
+from ..models import build_model
+from ..optim import make_optimizer
+from ..utils import PhaseTimers, flatten_params
+from .comm import Comm
+from .wire import WireCodec
+
+
+class PSTrainer:
+    def __init__(
+        self,
+        model_name: str,
+        codec: Codec,
+        comm: Comm,
+        lr: float = 0.01,
+        momentum: float = 0.9,
+        weight_decay: float = 0.0,
+        optimizer: str = "sgd",
+        lr_shrinkage: float = 0.95,
+        shrink_freq: int = 50,
+        num_classes: int = 10,
+        in_channels: int = 3,
+        dedicated_ps: bool = False,
+        device: Optional[torch.device] = None,
+        seed: Optional[int] = None,
+        checkpoint_freq: int = 0,
+        train_dir: str = "output/models/",
+    ):
+        self.comm = comm
+        self.device = device or comm.device
+        if seed is not None:
+            torch.manual_seed(seed)
+        self.model = build_model(model_name, num_classes, in_channels).to(self.device)
+        self.flat, self.params = flatten_params(self.model)
+        self.wc = WireCodec(codec, self.params, self.device)
+        self.codec = codec
+        self.loss_fn = nn.CrossEntropyLoss()
+
+        self.dedicated_ps = dedicated_ps and comm.world > 1
+        self.is_master = comm.rank == 0
+        self.is_worker = (not self.dedicated_ps) or comm.rank > 0
+        self.num_workers = comm.world - 1 if self.dedicated_ps else comm.world
+
+        self.wire = torch.zeros(
+            self.wc.total_words, dtype=torch.float32, device=self.device
+        )
+        self.gather_buf = (
+            torch.zeros(
+                max(1, comm.world),
+                self.wc.total_words,
+                dtype=torch.float32,
+                device=self.device,
+            )
+            if self.is_master and not self.wc.reducible
+            else None
+        )
+        self.agg = (
+            torch.zeros_like(self.flat)
+            if self.is_master and not self.wc.reducible
+            else None
+        )
+        self.opt = (
+            make_optimizer(
+                optimizer, self.flat, lr=lr, momentum=momentum, weight_decay=weight_decay
+            )
+            if self.is_master
+            else None
+        )
+        self.lr = lr
+        self.lr_shrinkage = lr_shrinkage
+        self.shrink_freq = shrink_freq
+        self.step_num = 0
+        self.checkpoint_freq = checkpoint_freq
+        self.train_dir = train_dir
+        self.timers = PhaseTimers()
+        self.last_loss = float("nan")
+
+        # make every rank start from rank-0's init
+        self.comm.broadcast(self.flat, src=0)
+
+    # -----------------------------------------------------------------
+    def train_step(self, x: torch.Tensor, y: torch.Tensor) -> float:
+        t = self.timers
+        with t.phase("fetch"):
+            self.comm.broadcast(self.flat, src=0)
+
+        if self.is_worker:
+            with t.phase("comp"):
+                self.model.train()
+                for p in self.params:
+                    if p.grad is not None:
+                        p.grad.detach_()
+                        p.grad.zero_()
+                out = self.model(x)
+                loss = self.loss_fn(out, y)
+                loss.backward()
+                self.last_loss = float(loss.detach())
+            with t.phase("encode"):
+                used = self.wc.encode_all(self.wire)
+                t.add_scalar("msg_bytes", 4.0 * used)
+
+        with t.phase("comm"):
+            if self.wc.reducible:
+                self.comm.reduce_sum(self.wire, dst=0)
+            else:
+                self.comm.gather(self.wire, self.gather_buf, dst=0)
+
+        if self.is_master:
+            with t.phase("decode"):
+                if self.wc.reducible:
+                    grad_flat = self.wire
+                else:
+                    self.agg.zero_()
+                    rows = self.gather_buf[: max(1, self.comm.world)]
+                    self.wc.decode_all(rows, self.agg)
+                    grad_flat = self.agg
+            with t.phase("apply"):
+                self.opt.lr = self.lr
+                self._apply(grad_flat)
+            if self.wc.reducible:
+                # the summed wire doubles as scratch; zero for next step
+                self.wire.zero_()
+
+        self.step_num += 1
+        if self.step_num % self.shrink_freq == 0:
+            self.lr *= self.lr_shrinkage
+        if (
+            self.is_master
+            and self.checkpoint_freq > 0
+            and self.step_num % self.checkpoint_freq == 0
+        ):
+            self.save_checkpoint()
+        return self.last_loss
+
+    def _apply(self, grad_flat: torch.Tensor) -> None:
+        scale = 1.0 / max(1, self.num_workers)
+        if self.flat.is_cuda and type(self.opt).__name__ == "ExternalSGD":
+            from ..ops import optim_ops
+
+            optim_ops.fused_sgd(
+                self.flat,
+                grad_flat,
+                self.opt.buf,
+                lr=self.opt.lr,
+                momentum=self.opt.momentum,
+                weight_decay=self.opt.weight_decay,
+                nesterov=self.opt.nesterov,
+                dampening=self.opt.dampening,
+                grad_scale=scale,
+            )
+        else:
+            if scale != 1.0:
+                grad_flat = grad_flat * scale
+            self.opt.step(grad_flat)
+
+    # -----------------------------------------------------------------
+    @torch.no_grad()
+    def evaluate(self, loader, max_batches: int = 0) -> dict:
+        """On-worker eval mirroring _evaluate_model
+        (distributed_worker.py:344-370): loss + Prec@1/Prec@5."""
+        from ..utils import accuracy
+
+        self.model.eval()
+        tot, loss_sum, p1_sum, p5_sum, nb = 0, 0.0, 0.0, 0.0, 0
+        for i, (x, y) in enumerate(loader):
+            if max_batches and i >= max_batches:
+                break
+            x, y = x.to(self.device), y.to(self.device)
+            out = self.model(x)
+            loss_sum += float(self.loss_fn(out, y)) * y.numel()
+            k5 = min(5, out.shape[1])
+            p1, p5 = accuracy(out, y, topk=(1, k5))
+            p1_sum += p1 * y.numel()
+            p5_sum += p5 * y.numel()
+            tot += y.numel()
+            nb += 1
+        self.model.train()
+        if tot == 0:
+            return {"loss": float("nan"), "prec1": 0.0, "prec5": 0.0}
+        return {"loss": loss_sum / tot, "prec1": p1_sum / tot, "prec5": p5_sum / tot}
+
+    def save_checkpoint(self, path: Optional[str] = None) -> str:
+        """torch.save to train_dir/model_step_<N> (reference
+        sync_replicas_master_nn.py:331-336) plus optimizer/step state so a
+        run can actually RESUME (the reference has no resume path)."""
+        os.makedirs(self.train_dir, exist_ok=True)
+        path = path or os.path.join(self.train_dir, f"model_step_{self.step_num}")
+        torch.save(
+            {
+                "step": self.step_num,
+                "lr": self.lr,
+                "model": self.model.state_dict(),
+                "optimizer": self.opt.state_dict() if self.opt else None,
+            },
+            path,
+        )
+        return path
+
+    def load_checkpoint(self, path: str) -> None:
+        ckpt = torch.load(path, map_location=self.device, weights_only=False)
+        self.model.load_state_dict(ckpt["model"])
+        # state_dict load replaced param storages; re-flatten into our buffer
+        self.flat, self.params = flatten_params(self.model)
+        self.wc = WireCodec(self.codec, self.params, self.device)
+        self.step_num = ckpt["step"]
+        self.lr = ckpt["lr"]
+        if self.opt is not None and ckpt.get("optimizer") is not None:
+            self.opt.p = self.flat
+            self.opt.load_state_dict(ckpt["optimizer"])
