@@ -1,0 +1,149 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ResNet-18 / CIFAR-10-shape PS training with SVD
+rank-3 atomic sparsification (BASELINE.json headline config).
+
+    python bench.py --gpus N --steps K --warmup W
+
+For N > 1 the driver launches this via torch.distributed.run (one rank per
+GPU over RCCL); rank 0 is the parameter server colocated with a worker.
+Synthetic CIFAR-10-shaped data, random-init weights, fp32 compute (the
+reference computes fp32; BASELINE names no lower dtype).  Reports the
+whole-job aggregate images/sec plus the reference's own counters
+(iters/sec, gradient MB/step/worker).
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch-size", type=int, default=128)
+    p.add_argument("--network", type=str, default="ResNet18")
+    p.add_argument("--dataset", type=str, default="cifar10")
+    p.add_argument("--code", type=str, default="svd")
+    p.add_argument("--svd-rank", type=int, default=3)
+    p.add_argument("--quantization-level", type=int, default=4)
+    p.add_argument("--bucket-size", type=int, default=512)
+    p.add_argument("--dedicated-ps", action="store_true", default=False)
+    p.add_argument("--cpu", action="store_true", default=False)
+    a = p.parse_args()
+
+    from atomo_amd.codings import make_codec
+    from atomo_amd.data import dataset_spec, make_loaders
+    from atomo_amd.parallel import Comm, PSTrainer
+
+    world_env = int(os.environ.get("WORLD_SIZE", "1"))
+    n_gpus = max(a.gpus, world_env)
+    device = torch.device("cpu") if (a.cpu or not torch.cuda.is_available()) else None
+
+    comm = Comm(device=device)
+    device = comm.device if device is None else device
+    codec = make_codec(
+        a.code,
+        rank=a.svd_rank,
+        quantization_level=a.quantization_level,
+        bucket_size=a.bucket_size,
+    )
+    spec = dataset_spec(a.dataset)
+    trainer = PSTrainer(
+        model_name=a.network,
+        codec=codec,
+        comm=comm,
+        lr=0.01,
+        momentum=0.9,
+        lr_shrinkage=1.0,  # fixed lr inside the timed window
+        num_classes=spec["classes"],
+        in_channels=spec["shape"][0],
+        dedicated_ps=a.dedicated_ps,
+        seed=42,
+        device=device,
+    )
+    train, _ = make_loaders(a.dataset, a.batch_size, a.batch_size, device,
+                            seed=123 + comm.rank)
+
+    def sync():
+        comm.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+
+    it = iter_cycle(train)
+    for _ in range(a.warmup):
+        x, y = next(it)
+        trainer.train_step(x, y)
+
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(a.steps):
+        x, y = next(it)
+        trainer.train_step(x, y)
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if comm.backend == "nccl" else "cpu")
+    if comm.world > 1:
+        import torch.distributed as dist
+
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    n_workers = trainer.num_workers
+    iters_per_sec = a.steps / elapsed
+    images_per_sec = iters_per_sec * a.batch_size * n_workers
+    msg_bytes = trainer.timers.scalars.get("msg_bytes", 0.0)
+    steps_counted = max(1, trainer.timers.counts.get("msg_bytes", 1))
+    grad_mb_per_step = (msg_bytes / steps_counted) / 1e6
+
+    if comm.rank == 0:
+        print(
+            json.dumps(
+                {
+                    "metric": "images/sec (ResNet-18 SVD-r=3 PS data-parallel)",
+                    "value": images_per_sec,
+                    "unit": "images/s",
+                    "n_gpus": n_gpus,
+                    "steps": a.steps,
+                    "warmup": a.warmup,
+                    "ms_per_step": 1e3 * elapsed / a.steps,
+                    "higher_is_better": True,
+                    "scaling": "weak",
+                    "vs_baseline": None,
+                    "dtype": "fp32",
+                    "data": "synthetic (CIFAR-10-shaped random, random-init weights)",
+                    "iters_per_sec": iters_per_sec,
+                    "grad_mb_per_step_per_worker": grad_mb_per_step,
+                    "config": {
+                        "model": a.network,
+                        "global_batch": a.batch_size * n_workers,
+                        "batch_per_worker": a.batch_size,
+                        "input": list(spec["shape"]),
+                        "code": a.code,
+                        "svd_rank": a.svd_rank,
+                        "parallelism": f"ps-dp{n_gpus}"
+                        + ("-dedicated" if a.dedicated_ps else "-colocated"),
+                    },
+                }
+            ),
+            flush=True,
+        )
+    comm.barrier()
+    comm.close()
+
+
+def iter_cycle(loader):
+    while True:
+        for batch in loader:
+            yield batch
+
+
+if __name__ == "__main__":
+    main()

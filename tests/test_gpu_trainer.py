@@ -1,0 +1,118 @@
+"""End-to-end GPU PS loop on one MI355X (world=1 self-PS): every codec,
+HIP kernels in the loop."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    return torch.device("cuda:0")
+
+
+def _trainer(code, dev, model="ResNet18", **kw):
+    from atomo_amd.codings import make_codec
+    from atomo_amd.parallel import Comm, PSTrainer
+
+    comm = Comm(device=dev)
+    codec = make_codec(code, rank=3, quantization_level=4, bucket_size=512)
+    return PSTrainer(
+        model_name=model,
+        codec=codec,
+        comm=comm,
+        lr=0.02,
+        momentum=0.9,
+        num_classes=10,
+        in_channels=3,
+        seed=11,
+        device=dev,
+        **kw,
+    )
+
+
+@pytest.mark.parametrize("code", ["sgd", "svd", "qsgd"])
+def test_gpu_loop(code, dev):
+    from atomo_amd.data import make_loaders
+
+    trainer = _trainer(code, dev)
+    train, _ = make_loaders("cifar10", 64, 64, dev, seed=3)
+    losses = []
+    for i, (x, y) in enumerate(train):
+        losses.append(trainer.train_step(x, y))
+        if i >= 11:
+            break
+    assert all(not math.isnan(l) for l in losses)
+    assert losses[-1] < losses[0], losses
+    assert torch.isfinite(trainer.flat).all()
+
+
+def test_gpu_raw_equals_plain_sgd(dev):
+    """Raw codec on GPU (fused kernels) == torch.optim.SGD trajectory."""
+    from atomo_amd.data import make_loaders
+    from atomo_amd.models import build_model
+    from atomo_amd.utils import flatten_params
+
+    torch.manual_seed(11)
+    trainer = _trainer("sgd", dev, model="LeNet")
+    trainer.model.load_state_dict(trainer.model.state_dict())
+
+    ref = build_model("LeNet", 10, 3).to(dev)
+    ref.load_state_dict(trainer.model.state_dict())
+    ref_flat, ref_params = flatten_params(ref)
+    opt = torch.optim.SGD(ref_params, lr=0.02, momentum=0.9)
+    loss_fn = torch.nn.CrossEntropyLoss()
+
+    train, _ = make_loaders("mnist", 32, 32, dev, seed=4)
+    # LeNet is 1-channel; rebuild with mnist shapes
+    trainer2 = None
+    from atomo_amd.codings import make_codec
+    from atomo_amd.parallel import Comm, PSTrainer
+
+    comm = Comm(device=dev)
+    torch.manual_seed(21)
+    trainer2 = PSTrainer(
+        model_name="LeNet", codec=make_codec("sgd"), comm=comm, lr=0.02,
+        momentum=0.9, num_classes=10, in_channels=1, seed=21, device=dev,
+    )
+    ref = build_model("LeNet", 10, 1).to(dev)
+    ref.load_state_dict(trainer2.model.state_dict())
+    ref_flat, ref_params = flatten_params(ref)
+    opt = torch.optim.SGD(ref_params, lr=0.02, momentum=0.9)
+    for i, (x, y) in enumerate(train):
+        trainer2.train_step(x, y)
+        opt.zero_grad()
+        loss_fn(ref(x), y).backward()
+        opt.step()
+        if i >= 4:
+            break
+    diff = (trainer2.flat - ref_flat).abs().max().item()
+    assert diff < 1e-4, diff
+
+
+def test_gpu_svd_wire_unbiased(dev):
+    """Whole wire round trip (encode_into + decode_acc kernel) is unbiased."""
+    from atomo_amd.codings import SVDCodec
+
+    torch.manual_seed(12)
+    g = torch.randn(64, 16, 3, 3, device=dev)
+    codec = SVDCodec(rank=3, backend="gram")
+    acc = torch.zeros(g.numel(), device=dev)
+    spec = codec.build_specs([list(g.shape)])[0]
+    wire = torch.zeros(spec.wire_words, device=dev)
+    n = 300
+    from atomo_amd.ops import svd_ops
+
+    meta = spec.meta
+    for _ in range(n):
+        codec.encode_into(g, wire, spec)
+        svd_ops.decode_acc(
+            wire.view(1, -1), acc.view(meta["m"], meta["n"]), meta["m"],
+            meta["n"], meta["r_max"],
+        )
+    rel = ((acc / n) - g.reshape(-1)).norm() / g.norm()
+    assert rel < 0.35, rel
