@@ -74,9 +74,23 @@ class PSTrainer:
         self.is_worker = (not self.dedicated_ps) or comm.rank > 0
         self.num_workers = comm.world - 1 if self.dedicated_ps else comm.world
 
-        self.wire = torch.zeros(
-            self.wc.total_words, dtype=torch.float32, device=self.device
-        )
+        # gradients live in one flat buffer (views as p.grad): zeroing is one
+        # kernel, and for the raw codec the wire IS the grad buffer (layer
+        # order and sizes coincide), so encode is free and RCCL reduces the
+        # gradients in place.
+        self.flat_grad = torch.zeros_like(self.flat)
+        off = 0
+        for p in self.params:
+            n = p.numel()
+            p.grad = self.flat_grad[off : off + n].view_as(p)
+            off += n
+        if self.wc.reducible:
+            assert self.wc.total_words == self.flat.numel()
+            self.wire = self.flat_grad
+        else:
+            self.wire = torch.zeros(
+                self.wc.total_words, dtype=torch.float32, device=self.device
+            )
         self.gather_buf = (
             torch.zeros(
                 max(1, comm.world),
@@ -120,17 +134,19 @@ class PSTrainer:
         if self.is_worker:
             with t.phase("comp"):
                 self.model.train()
-                for p in self.params:
-                    if p.grad is not None:
-                        p.grad.detach_()
-                        p.grad.zero_()
+                self.flat_grad.zero_()
                 out = self.model(x)
                 loss = self.loss_fn(out, y)
                 loss.backward()
                 self.last_loss = float(loss.detach())
             with t.phase("encode"):
-                used = self.wc.encode_all(self.wire)
+                if self.wc.reducible:
+                    used = self.wc.total_words  # wire aliases flat_grad
+                else:
+                    used = self.wc.encode_all(self.wire)
                 t.add_scalar("msg_bytes", 4.0 * used)
+        elif self.wc.reducible:
+            self.flat_grad.zero_()  # dedicated PS contributes zeros to the sum
 
         with t.phase("comm"):
             if self.wc.reducible:
@@ -150,9 +166,6 @@ class PSTrainer:
             with t.phase("apply"):
                 self.opt.lr = self.lr
                 self._apply(grad_flat)
-            if self.wc.reducible:
-                # the summed wire doubles as scratch; zero for next step
-                self.wire.zero_()
 
         self.step_num += 1
         if self.step_num % self.shrink_freq == 0:
@@ -235,6 +248,14 @@ class PSTrainer:
         # state_dict load replaced param storages; re-flatten into our buffer
         self.flat, self.params = flatten_params(self.model)
         self.wc = WireCodec(self.codec, self.params, self.device)
+        self.flat_grad = torch.zeros_like(self.flat)
+        off = 0
+        for p in self.params:
+            n = p.numel()
+            p.grad = self.flat_grad[off : off + n].view_as(p)
+            off += n
+        if self.wc.reducible:
+            self.wire = self.flat_grad
         self.step_num = ckpt["step"]
         self.lr = ckpt["lr"]
         if self.opt is not None and ckpt.get("optimizer") is not None:

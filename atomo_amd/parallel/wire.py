@@ -29,6 +29,16 @@ class WireCodec:
         self.device = device
         self.specs = codec.build_specs([list(p.shape) for p in params])
         self.total_words = sum(s.wire_words for s in self.specs)
+        self._batched_encoder = None
+        if (
+            device.type == "cuda"
+            and isinstance(codec, SVDCodec)
+            and codec.compress
+            and codec.backend in ("auto", "gram")
+        ):
+            from .svd_encoder import BatchedSVDEncoder
+
+            self._batched_encoder = BatchedSVDEncoder(codec, self.specs, device)
         # flat-parameter offsets (agg buffer layout == flat param layout)
         self.param_offsets = []
         off = 0
@@ -53,9 +63,13 @@ class WireCodec:
     def encode_all(self, wire: torch.Tensor) -> int:
         """Encode every parameter's .grad into ``wire``; returns fp32 words
         actually used (the Msg bytes counter)."""
+        grads = [
+            p.grad if p.grad is not None else torch.zeros_like(p) for p in self.params
+        ]
+        if self._batched_encoder is not None:
+            return self._batched_encoder.encode_all(grads, wire)
         used = 0
-        for p, spec in zip(self.params, self.specs):
-            grad = p.grad if p.grad is not None else torch.zeros_like(p)
+        for grad, spec in zip(grads, self.specs):
             region = wire[spec.wire_offset : spec.wire_offset + spec.wire_words]
             used += self.codec.encode_into(grad, region, spec)
         return used
