@@ -21,10 +21,14 @@ codings/svd.py:49-117)."""
 
 from __future__ import annotations
 
+import os
+import time
 from collections import defaultdict
 from typing import List
 
 import torch
+
+_TRACE = os.environ.get("ATOMO_TRACE_ENCODER", "") not in ("", "0")
 
 from ..codings.base import LayerSpec
 from ..codings.svd import SVDCodec, sample_svd
@@ -35,6 +39,10 @@ class BatchedSVDEncoder:
         self.codec = codec
         self.device = device
         self.specs = [s for s in specs]
+        self.eigh_on_device = (
+            device.type == "cuda"
+            and os.environ.get("ATOMO_EIGH_DEVICE", "0") not in ("", "0")
+        )
         # per-layer geometry
         self.small = []  # small dim (rank side)
         self.tall = []  # tall dim
@@ -78,6 +86,15 @@ class BatchedSVDEncoder:
 
     @torch.no_grad()
     def encode_all(self, grads: List[torch.Tensor], wire: torch.Tensor) -> int:
+        marks = [time.perf_counter()] if _TRACE else None
+
+        def mark(label):
+            if _TRACE:
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize()
+                marks.append(time.perf_counter())
+                print(f"[enc] {label}: {1e3*(marks[-1]-marks[-2]):.2f} ms", flush=True)
+
         specs = self.specs
         a2ds = [self._a2d(g, s) for g, s in zip(grads, specs)]
 
@@ -91,29 +108,56 @@ class BatchedSVDEncoder:
                 torch.mm(a.t(), a, out=gv)
             else:
                 torch.mm(a, a.t(), out=gv)
+        mark("A grams")
 
-        # ---- phase B: ONE sync, host eigensolves + sampling ------------
-        grams_host = self.grams.to("cpu", non_blocking=False)
+        # ---- phase B: eigensolves + ONE sync + sampling ----------------
         by_dim = defaultdict(list)
         for i in range(len(specs)):
             by_dim[self.small[i]].append(i)
         evecs_h, svals_h = {}, {}
-        for sm, idxs in by_dim.items():
-            gs = torch.stack(
-                [
-                    grams_host[
-                        self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
-                    ].view(sm, sm)
-                    for i in idxs
-                ]
-            ).to(torch.float64)
-            gs = 0.5 * (gs + gs.transpose(1, 2))  # symmetrize fp32 roundoff
-            evals, evecs = torch.linalg.eigh(gs)  # ascending
-            evals = evals.flip(1).clamp(min=0.0)
-            evecs = evecs.flip(2)
-            for j, i in enumerate(idxs):
-                svals_h[i] = evals[j].sqrt()
-                evecs_h[i] = evecs[j]
+        if self.eigh_on_device:
+            # batched hipSOLVER eigh per distinct small-dim, ONE D2H after
+            results = []
+            for sm, idxs in by_dim.items():
+                gs = torch.stack(
+                    [
+                        self.grams[
+                            self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
+                        ].view(sm, sm)
+                        for i in idxs
+                    ]
+                )
+                gs = 0.5 * (gs + gs.transpose(1, 2))
+                evals, evecs = torch.linalg.eigh(gs)
+                results.append((idxs, evals, evecs))
+            for idxs, evals, evecs in results:
+                evals_h = evals.to("cpu", torch.float64)
+                evecs_hh = evecs.to("cpu", torch.float64)
+                evals_h = evals_h.flip(1).clamp(min=0.0)
+                evecs_hh = evecs_hh.flip(2)
+                for j, i in enumerate(idxs):
+                    svals_h[i] = evals_h[j].sqrt()
+                    evecs_h[i] = evecs_hh[j]
+        else:
+            grams_host = self.grams.to("cpu", non_blocking=False)
+            mark("B d2h")
+            for sm, idxs in by_dim.items():
+                gs = torch.stack(
+                    [
+                        grams_host[
+                            self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
+                        ].view(sm, sm)
+                        for i in idxs
+                    ]
+                ).to(torch.float64)
+                gs = 0.5 * (gs + gs.transpose(1, 2))  # symmetrize fp32 roundoff
+                evals, evecs = torch.linalg.eigh(gs)  # ascending
+                evals = evals.flip(1).clamp(min=0.0)
+                evecs = evecs.flip(2)
+                for j, i in enumerate(idxs):
+                    svals_h[i] = evals[j].sqrt()
+                    evecs_h[i] = evecs[j]
+        mark("B eigh")
 
         used = 0
         plans = []
@@ -152,6 +196,7 @@ class BatchedSVDEncoder:
             stage[sc_off : sc_off + sm * r_hat] = sel_scaled.reshape(-1)
             plans.append((i, r_hat))
             used += 1 + r_hat * (spec.meta["m"] + spec.meta["n"] + 1)
+        mark("B sample+stage")
 
         # ---- phase C: one H2D + per-layer GEMMs into the wire ----------
         self.stage_dev.copy_(self.stage_host, non_blocking=True)
@@ -186,4 +231,5 @@ class BatchedSVDEncoder:
                 v_off = wo + 1 + r_max * (m + 1)
                 v_out = wire[v_off : v_off + r_hat * n].view(r_hat, n)
                 torch.mm(sel.t(), a, out=v_out)
+        mark("C h2d+gemms")
         return used

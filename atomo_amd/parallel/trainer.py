@@ -61,6 +61,11 @@ class PSTrainer:
     ):
         self.comm = comm
         self.device = device or comm.device
+        if self.device.type == "cuda":
+            # GPU ranks: host work is tiny batched eigensolves + sampling —
+            # hundreds of OpenMP threads (EPYC boxes) make a 6 ms LAPACK
+            # call take 600 ms.  Cap them.
+            torch.set_num_threads(min(8, os.cpu_count() or 8))
         if seed is not None:
             torch.manual_seed(seed)
         self.model = build_model(model_name, num_classes, in_channels).to(self.device)

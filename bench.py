@@ -34,6 +34,8 @@ def main():
     p.add_argument("--bucket-size", type=int, default=512)
     p.add_argument("--dedicated-ps", action="store_true", default=False)
     p.add_argument("--cpu", action="store_true", default=False)
+    p.add_argument("--phase-log", action="store_true", default=False,
+                   help="print per-phase timer breakdown to stderr")
     a = p.parse_args()
 
     from atomo_amd.codings import make_codec
@@ -80,12 +82,20 @@ def main():
         trainer.train_step(x, y)
 
     sync()
+    trainer.timers.reset()
+    if a.phase_log:
+        trainer.timers.sync_cuda = device.type == "cuda"
     t0 = time.perf_counter()
     for _ in range(a.steps):
         x, y = next(it)
         trainer.train_step(x, y)
     sync()
     elapsed = time.perf_counter() - t0
+    if a.phase_log:
+        import sys
+
+        print(json.dumps({"rank": comm.rank, **trainer.timers.summary()}),
+              file=sys.stderr, flush=True)
 
     # max over ranks
     t = torch.tensor([elapsed], dtype=torch.float64,
