@@ -18,6 +18,11 @@ void atomo_svd_decode_acc_launch(const float*, float*, int, int64_t, int, int,
                                  int, hipStream_t);
 void atomo_fused_sgd_launch(float*, const float*, float*, int64_t, float,
                             float, float, bool, float, float, hipStream_t);
+void atomo_batched_gram_launch(const float*, float*, const int64_t*,
+                               const int32_t*, int, hipStream_t);
+void atomo_batched_sel_launch(const float*, float*, const float*,
+                              const int64_t*, const int32_t*, int,
+                              hipStream_t);
 }
 
 namespace {
@@ -101,9 +106,45 @@ void fused_sgd(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
                          cur_stream());
 }
 
+void batched_gram(torch::Tensor flat, torch::Tensor grams, torch::Tensor desc,
+                  torch::Tensor work, int64_t n_tiles) {
+  check_f32_cuda(flat, "flat");
+  check_f32_cuda(grams, "grams");
+  TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kInt64 &&
+                  desc.is_contiguous(),
+              "desc must be contiguous cuda int64");
+  TORCH_CHECK(work.is_cuda() && work.scalar_type() == torch::kInt32 &&
+                  work.is_contiguous(),
+              "work must be contiguous cuda int32");
+  atomo_batched_gram_launch(flat.data_ptr<float>(), grams.data_ptr<float>(),
+                            desc.data_ptr<int64_t>(), work.data_ptr<int32_t>(),
+                            (int)n_tiles, cur_stream());
+}
+
+void batched_sel(torch::Tensor flat, torch::Tensor wire, torch::Tensor stage,
+                 torch::Tensor desc, torch::Tensor work, int64_t n_tiles) {
+  check_f32_cuda(flat, "flat");
+  check_f32_cuda(wire, "wire");
+  check_f32_cuda(stage, "stage");
+  TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kInt64 &&
+                  desc.is_contiguous(),
+              "desc must be contiguous cuda int64");
+  TORCH_CHECK(work.is_cuda() && work.scalar_type() == torch::kInt32 &&
+                  work.is_contiguous(),
+              "work must be contiguous cuda int32");
+  atomo_batched_sel_launch(flat.data_ptr<float>(), wire.data_ptr<float>(),
+                           stage.data_ptr<float>(), desc.data_ptr<int64_t>(),
+                           work.data_ptr<int32_t>(), (int)n_tiles,
+                           cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("batched_gram", &batched_gram,
+        "batched per-layer Gram matrices (small-dim <= 64)");
+  m.def("batched_sel", &batched_sel,
+        "batched selection GEMM + packet scatter into the wire");
   m.def("qsgd_pack", &qsgd_pack, "QSGD bucket quantize+pack (gfx950)");
   m.def("qsgd_unpack_acc", &qsgd_unpack_acc, "QSGD unpack+accumulate");
   m.def("svd_decode_acc", &svd_decode_acc,

@@ -35,7 +35,13 @@ from ..codings.svd import SVDCodec, sample_svd
 
 
 class BatchedSVDEncoder:
-    def __init__(self, codec: SVDCodec, specs: List[LayerSpec], device: torch.device):
+    def __init__(
+        self,
+        codec: SVDCodec,
+        specs: List[LayerSpec],
+        device: torch.device,
+        param_offsets: List[int] | None = None,
+    ):
         self.codec = codec
         self.device = device
         self.specs = [s for s in specs]
@@ -43,6 +49,7 @@ class BatchedSVDEncoder:
             device.type == "cuda"
             and os.environ.get("ATOMO_EIGH_DEVICE", "0") not in ("", "0")
         )
+        self.param_offsets = param_offsets
         # per-layer geometry
         self.small = []  # small dim (rank side)
         self.tall = []  # tall dim
@@ -74,6 +81,53 @@ class BatchedSVDEncoder:
             for s in specs
             if s.meta["padded"] != s.numel
         }
+        # ---- descriptor tables for the batched HIP kernels -------------
+        # (one launch for all small-dim<=64 layers; big/odd layers stay on
+        # rocBLAS GEMMs below)
+        self.kernel_set = set()
+        self.use_kernels = False
+        if device.type == "cuda" and param_offsets is not None:
+            from .. import ops
+
+            self.use_kernels = ops.have_ext()
+        if self.use_kernels:
+            GRAM_CHUNK, SEL_CHUNK = 256, 1024
+            desc_rows, gram_work, sel_work = [], [], []
+            for i, s in enumerate(specs):
+                m, n = s.meta["m"], s.meta["n"]
+                sm, tall = self.small[i], self.tall[i]
+                if (
+                    s.meta["padded"] != s.numel
+                    or sm > 64
+                    or s.meta["r_max"] > 16
+                ):
+                    continue
+                row = len(desc_rows)
+                desc_rows.append(
+                    [
+                        param_offsets[i],
+                        m,
+                        n,
+                        1 if self.m_is_tall[i] else 0,
+                        self.gram_offsets[i],
+                        s.wire_offset,
+                        self.stage_offsets[i],
+                        s.meta["r_max"],
+                    ]
+                )
+                self.kernel_set.add(i)
+                for c in range((tall + GRAM_CHUNK - 1) // GRAM_CHUNK):
+                    gram_work.append([row, c])
+                for c in range((tall + SEL_CHUNK - 1) // SEL_CHUNK):
+                    sel_work.append([row, c])
+            if desc_rows:
+                self.desc = torch.tensor(desc_rows, dtype=torch.int64, device=device)
+                self.gram_work = torch.tensor(
+                    gram_work, dtype=torch.int32, device=device
+                )
+                self.sel_work = torch.tensor(sel_work, dtype=torch.int32, device=device)
+            else:
+                self.use_kernels = False
 
     def _a2d(self, grad: torch.Tensor, spec: LayerSpec) -> torch.Tensor:
         m, n = spec.meta["m"], spec.meta["n"]
@@ -85,7 +139,14 @@ class BatchedSVDEncoder:
         return flat.view(m, n)
 
     @torch.no_grad()
-    def encode_all(self, grads: List[torch.Tensor], wire: torch.Tensor) -> int:
+    def encode_all(
+        self,
+        grads: List[torch.Tensor],
+        wire: torch.Tensor,
+        flat_grad: torch.Tensor | None = None,
+    ) -> int:
+        use_kernels = self.use_kernels and flat_grad is not None
+        kernel_set = self.kernel_set if use_kernels else set()
         marks = [time.perf_counter()] if _TRACE else None
 
         def mark(label):
@@ -99,7 +160,17 @@ class BatchedSVDEncoder:
         a2ds = [self._a2d(g, s) for g, s in zip(grads, specs)]
 
         # ---- phase A: Grams on device (async) --------------------------
+        if use_kernels:
+            from ..ops import ext
+
+            self.grams.zero_()
+            ext().batched_gram(
+                flat_grad, self.grams, self.desc, self.gram_work,
+                self.gram_work.shape[0],
+            )
         for i, (a, s) in enumerate(zip(a2ds, specs)):
+            if i in kernel_set:
+                continue
             sm = self.small[i]
             gv = self.grams[self.gram_offsets[i] : self.gram_offsets[i] + sm * sm].view(
                 sm, sm
@@ -159,6 +230,38 @@ class BatchedSVDEncoder:
                     evecs_h[i] = evecs[j]
         mark("B eigh")
 
+        # vectorized importance sampling: ONE rand over every layer's probs
+        samples = {}
+        if self.codec.random_sample:
+            rank = self.codec.rank
+            probs_list = []
+            for i in range(len(specs)):
+                s = svals_h[i].float()
+                if s.numel() == 0 or float(s[0]) < 1e-6:
+                    probs_list.append(torch.zeros(0))
+                    samples[i] = (torch.tensor([0]), torch.tensor([1.0]))
+                    continue
+                p = (s / s[0]) if rank == 0 else (rank * s / s.sum())
+                probs_list.append(p.clamp(max=1.0))
+            cat = torch.cat(probs_list) if probs_list else torch.zeros(0)
+            draws = torch.rand(cat.shape, generator=self.codec.generator) < cat
+            off = 0
+            for i in range(len(specs)):
+                if i in samples:
+                    continue
+                p = probs_list[i]
+                k = p.numel()
+                d = draws[off : off + k]
+                off += k
+                idx = d.nonzero(as_tuple=False).flatten()
+                if idx.numel() == 0:  # rare: redraw this layer alone
+                    idx, pr = sample_svd(
+                        svals_h[i].float(), rank=rank, generator=self.codec.generator
+                    )
+                    samples[i] = (idx, pr)
+                else:
+                    samples[i] = (idx, p[idx])
+
         used = 0
         plans = []
         stage = self.stage_host
@@ -168,9 +271,7 @@ class BatchedSVDEncoder:
             s64 = svals_h[i]
             v64 = evecs_h[i]
             if self.codec.random_sample:
-                idx, probs = sample_svd(
-                    s64.float(), rank=self.codec.rank, generator=self.codec.generator
-                )
+                idx, probs = samples[i]
                 if idx.numel() > r_max:
                     self.codec.overflow_count += 1
                     idx, probs = idx[:r_max], probs[:r_max]
@@ -198,10 +299,19 @@ class BatchedSVDEncoder:
             used += 1 + r_hat * (spec.meta["m"] + spec.meta["n"] + 1)
         mark("B sample+stage")
 
-        # ---- phase C: one H2D + per-layer GEMMs into the wire ----------
+        # ---- phase C: one H2D + one batched kernel (+ rocBLAS leftovers)
         self.stage_dev.copy_(self.stage_host, non_blocking=True)
         sd = self.stage_dev
+        if use_kernels:
+            from ..ops import ext
+
+            ext().batched_sel(
+                flat_grad, wire, sd, self.desc, self.sel_work,
+                self.sel_work.shape[0],
+            )
         for i, r_hat in plans:
+            if i in kernel_set:
+                continue
             spec = specs[i]
             m, n, r_max = spec.meta["m"], spec.meta["n"], spec.meta["r_max"]
             sm, tall = self.small[i], self.tall[i]

@@ -103,7 +103,7 @@ class PSTrainer:
                 dtype=torch.float32,
                 device=self.device,
             )
-            if self.is_master and not self.wc.reducible
+            if self.is_master and not self.wc.reducible and comm.world > 1
             else None
         )
         self.agg = (
@@ -148,7 +148,7 @@ class PSTrainer:
                 if self.wc.reducible:
                     used = self.wc.total_words  # wire aliases flat_grad
                 else:
-                    used = self.wc.encode_all(self.wire)
+                    used = self.wc.encode_all(self.wire, flat_grad=self.flat_grad)
                 t.add_scalar("msg_bytes", 4.0 * used)
         elif self.wc.reducible:
             self.flat_grad.zero_()  # dedicated PS contributes zeros to the sum
@@ -156,7 +156,7 @@ class PSTrainer:
         with t.phase("comm"):
             if self.wc.reducible:
                 self.comm.reduce_sum(self.wire, dst=0)
-            else:
+            elif self.comm.world > 1:
                 self.comm.gather(self.wire, self.gather_buf, dst=0)
 
         if self.is_master:
@@ -165,7 +165,11 @@ class PSTrainer:
                     grad_flat = self.wire
                 else:
                     self.agg.zero_()
-                    rows = self.gather_buf[: max(1, self.comm.world)]
+                    rows = (
+                        self.gather_buf
+                        if self.comm.world > 1
+                        else self.wire.view(1, -1)
+                    )
                     self.wc.decode_all(rows, self.agg)
                     grad_flat = self.agg
             with t.phase("apply"):

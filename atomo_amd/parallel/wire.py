@@ -30,6 +30,13 @@ class WireCodec:
         self.specs = codec.build_specs([list(p.shape) for p in params])
         self.total_words = sum(s.wire_words for s in self.specs)
         self._batched_encoder = None
+        # flat-parameter offsets (agg buffer layout == flat param layout)
+        self.param_offsets = []
+        off = 0
+        for p in params:
+            self.param_offsets.append(off)
+            off += p.numel()
+        self.total_params = off
         if (
             device.type == "cuda"
             and isinstance(codec, SVDCodec)
@@ -38,14 +45,9 @@ class WireCodec:
         ):
             from .svd_encoder import BatchedSVDEncoder
 
-            self._batched_encoder = BatchedSVDEncoder(codec, self.specs, device)
-        # flat-parameter offsets (agg buffer layout == flat param layout)
-        self.param_offsets = []
-        off = 0
-        for p in params:
-            self.param_offsets.append(off)
-            off += p.numel()
-        self.total_params = off
+            self._batched_encoder = BatchedSVDEncoder(
+                codec, self.specs, device, param_offsets=self.param_offsets
+            )
         # scratch for layers whose 2-D fold is zero-padded (odd 1-D sizes)
         self._pad_scratch = {}
         if isinstance(codec, SVDCodec) and codec.compress:
@@ -60,14 +62,14 @@ class WireCodec:
         return getattr(self.codec, "reducible", False)
 
     # -- worker side -----------------------------------------------------
-    def encode_all(self, wire: torch.Tensor) -> int:
+    def encode_all(self, wire: torch.Tensor, flat_grad: torch.Tensor = None) -> int:
         """Encode every parameter's .grad into ``wire``; returns fp32 words
         actually used (the Msg bytes counter)."""
         grads = [
             p.grad if p.grad is not None else torch.zeros_like(p) for p in self.params
         ]
         if self._batched_encoder is not None:
-            return self._batched_encoder.encode_all(grads, wire)
+            return self._batched_encoder.encode_all(grads, wire, flat_grad=flat_grad)
         used = 0
         for grad, spec in zip(grads, self.specs):
             region = wire[spec.wire_offset : spec.wire_offset + spec.wire_words]

@@ -21,6 +21,7 @@ ext = CUDAExtension(
     sources=[
         "atomo_amd/ops/csrc/bindings.cpp",
         "atomo_amd/ops/csrc/atomo_kernels.hip",
+        "atomo_amd/ops/csrc/svd_batched.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
