@@ -23,6 +23,11 @@ void atomo_batched_gram_launch(const float*, float*, const int64_t*,
 void atomo_batched_sel_launch(const float*, float*, const float*,
                               const int64_t*, const int32_t*, int,
                               hipStream_t);
+void atomo_jacobi_eigh_launch(float*, float*, const int64_t*, const int64_t*,
+                              int, hipStream_t);
+void atomo_build_stage_launch(const float*, const float*, const float*,
+                              float*, const int64_t*, const int64_t*, int,
+                              hipStream_t);
 }
 
 namespace {
@@ -138,9 +143,43 @@ void batched_sel(torch::Tensor flat, torch::Tensor wire, torch::Tensor stage,
                            cur_stream());
 }
 
+void jacobi_eigh(torch::Tensor grams, torch::Tensor evals, torch::Tensor desc,
+                 torch::Tensor eval_offs, int64_t n_layers) {
+  check_f32_cuda(grams, "grams");
+  check_f32_cuda(evals, "evals");
+  TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kInt64 &&
+                  desc.is_contiguous(),
+              "desc must be contiguous cuda int64");
+  TORCH_CHECK(eval_offs.is_cuda() && eval_offs.scalar_type() == torch::kInt64,
+              "eval_offs must be cuda int64");
+  atomo_jacobi_eigh_launch(grams.data_ptr<float>(), evals.data_ptr<float>(),
+                           desc.data_ptr<int64_t>(),
+                           eval_offs.data_ptr<int64_t>(), (int)n_layers,
+                           cur_stream());
+}
+
+void build_stage(torch::Tensor evecs, torch::Tensor evals,
+                 torch::Tensor sel_table, torch::Tensor stage,
+                 torch::Tensor desc, torch::Tensor eval_offs,
+                 int64_t n_layers) {
+  check_f32_cuda(evecs, "evecs");
+  check_f32_cuda(evals, "evals");
+  check_f32_cuda(sel_table, "sel_table");
+  check_f32_cuda(stage, "stage");
+  atomo_build_stage_launch(
+      evecs.data_ptr<float>(), evals.data_ptr<float>(),
+      sel_table.data_ptr<float>(), stage.data_ptr<float>(),
+      desc.data_ptr<int64_t>(), eval_offs.data_ptr<int64_t>(), (int)n_layers,
+      cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("jacobi_eigh", &jacobi_eigh,
+        "batched parallel-Jacobi symmetric eigensolver (sm <= 64)");
+  m.def("build_stage", &build_stage,
+        "gather sampled atoms into the staged wire factors");
   m.def("batched_gram", &batched_gram,
         "batched per-layer Gram matrices (small-dim <= 64)");
   m.def("batched_sel", &batched_sel,

@@ -1,37 +1,43 @@
 """Batched SVD wire encoder — ONE host sync per step for the whole model.
 
-The naive per-layer path costs ~60 round trips per step (factorize, sample,
-write).  This encoder restructures the work MI355X-first:
+The naive per-layer path costs ~60 host round trips per step (factorize,
+sample, write).  This encoder restructures the work MI355X-first; layers
+split into two classes:
 
-  phase A (device, async): per-layer Gram matrices G_l of the SMALL dimension
-      (A^T A when m >= n, A A^T otherwise) via rocBLAS GEMMs into one
-      concatenated device buffer.  Reading every gradient once (~45 MB for
-      ResNet-18) at HBM speed; the O(m n^2) flops ride on MFMA.
-  phase B (one D2H sync): all Grams to host; fp64 eigensolves BATCHED per
-      distinct small-dim (n <= ~512 for every BASELINE model config);
-      importance-sample every layer (Bernoulli, p_i = min(1, r*s_i/sum s)).
-  phase C (host -> device, async): one pinned-staging H2D of the selection
-      factors; per-layer GEMM writes the tall factor STRAIGHT into its wire
-      region (out= a view, no transpose copies); small factors / s / header
-      are sliced from the staged tensor.
+KERNEL layers (small-dim <= 64, even fold — every conv/BN/bias and most fc):
+  entirely on device via the hand-written gfx950 kernels
+  (ops/csrc/svd_batched.hip, ops/csrc/jacobi_eigh.hip):
+    batched_gram -> jacobi_eigh (parallel cyclic Jacobi, one wave64 per
+    matrix) -> [evals to host: the ONLY sync] -> host Bernoulli sampling
+    (vectorized across layers) -> sel_table H2D -> build_stage ->
+    batched_sel writes the wire packets.
+HOST layers (small-dim > 64: a few 1x1-conv/fc folds, odd-padded 1-D):
+  Gram via rocBLAS, fp64/fp32 LAPACK eigh on a thread pool, staged
+  selection factors H2D, rocBLAS selection GEMMs straight into the wire.
 
-Semantics identical to SVDCodec.encode_into (same wire layout, same sampler,
-same unbiasedness invariant E[sum s_i/p_i u_i v_i^T] = grad; reference
-codings/svd.py:49-117)."""
+Semantics identical to SVDCodec.encode_into (same wire layout, same
+sampler, same unbiasedness invariant E[sum s_i/p_i u_i v_i^T] = grad;
+reference codings/svd.py:49-117)."""
 
 from __future__ import annotations
 
 import os
 import time
 from collections import defaultdict
-from typing import List
+from concurrent.futures import ThreadPoolExecutor
+from typing import List, Optional
 
 import torch
 
-_TRACE = os.environ.get("ATOMO_TRACE_ENCODER", "") not in ("", "0")
-
 from ..codings.base import LayerSpec
 from ..codings.svd import SVDCodec, sample_svd
+
+_TRACE = os.environ.get("ATOMO_TRACE_ENCODER", "") not in ("", "0")
+
+GRAM_CHUNK = 256
+SEL_CHUNK = 1024
+SEL_ROW = 33  # [r_hat | idx*16 | probs*16]
+R_CAP = 16
 
 
 class BatchedSVDEncoder:
@@ -40,20 +46,16 @@ class BatchedSVDEncoder:
         codec: SVDCodec,
         specs: List[LayerSpec],
         device: torch.device,
-        param_offsets: List[int] | None = None,
+        param_offsets: Optional[List[int]] = None,
     ):
         self.codec = codec
         self.device = device
-        self.specs = [s for s in specs]
-        self.eigh_on_device = (
-            device.type == "cuda"
-            and os.environ.get("ATOMO_EIGH_DEVICE", "0") not in ("", "0")
-        )
+        self.specs = list(specs)
         self.param_offsets = param_offsets
+        self._pool = ThreadPoolExecutor(max_workers=4)
+
         # per-layer geometry
-        self.small = []  # small dim (rank side)
-        self.tall = []  # tall dim
-        self.m_is_tall = []  # True when m >= n (u is the tall factor)
+        self.small, self.tall, self.m_is_tall = [], [], []
         gram_off, offs = 0, []
         for s in specs:
             m, n = s.meta["m"], s.meta["n"]
@@ -65,8 +67,8 @@ class BatchedSVDEncoder:
             gram_off += sm * sm
         self.gram_offsets = offs
         self.grams = torch.zeros(gram_off, dtype=torch.float32, device=device)
-        # staging: per-layer [r_hat | s_wire (r_max) | small_factor (sm*r_max)
-        #                      | sel_scaled (sm*r_max)]
+
+        # staging: per-layer [r_hat | s_wire(r_max) | facT(r_max*sm) | sel(sm*r_max)]
         st_off, st_offs = 0, []
         for i, s in enumerate(specs):
             st_offs.append(st_off)
@@ -75,32 +77,29 @@ class BatchedSVDEncoder:
         pin = device.type == "cuda"
         self.stage_host = torch.zeros(st_off, dtype=torch.float32, pin_memory=pin)
         self.stage_dev = torch.zeros(st_off, dtype=torch.float32, device=device)
+
         # scratch for odd-padded layers
         self._pad_scratch = {
             s.index: torch.zeros(s.meta["padded"], dtype=torch.float32, device=device)
             for s in specs
             if s.meta["padded"] != s.numel
         }
+
         # ---- descriptor tables for the batched HIP kernels -------------
-        # (one launch for all small-dim<=64 layers; big/odd layers stay on
-        # rocBLAS GEMMs below)
         self.kernel_set = set()
+        self.kernel_rows: List[int] = []  # desc row -> layer index
         self.use_kernels = False
         if device.type == "cuda" and param_offsets is not None:
             from .. import ops
 
             self.use_kernels = ops.have_ext()
         if self.use_kernels:
-            GRAM_CHUNK, SEL_CHUNK = 256, 1024
-            desc_rows, gram_work, sel_work = [], [], []
+            desc_rows, gram_work, sel_work, eval_offs = [], [], [], []
+            ev_off = 0
             for i, s in enumerate(specs):
                 m, n = s.meta["m"], s.meta["n"]
                 sm, tall = self.small[i], self.tall[i]
-                if (
-                    s.meta["padded"] != s.numel
-                    or sm > 64
-                    or s.meta["r_max"] > 16
-                ):
+                if s.meta["padded"] != s.numel or sm > 64 or s.meta["r_max"] > R_CAP:
                     continue
                 row = len(desc_rows)
                 desc_rows.append(
@@ -115,20 +114,38 @@ class BatchedSVDEncoder:
                         s.meta["r_max"],
                     ]
                 )
+                eval_offs.append(ev_off)
+                ev_off += sm
                 self.kernel_set.add(i)
+                self.kernel_rows.append(i)
                 for c in range((tall + GRAM_CHUNK - 1) // GRAM_CHUNK):
                     gram_work.append([row, c])
                 for c in range((tall + SEL_CHUNK - 1) // SEL_CHUNK):
                     sel_work.append([row, c])
             if desc_rows:
                 self.desc = torch.tensor(desc_rows, dtype=torch.int64, device=device)
-                self.gram_work = torch.tensor(
-                    gram_work, dtype=torch.int32, device=device
-                )
+                self.gram_work = torch.tensor(gram_work, dtype=torch.int32, device=device)
                 self.sel_work = torch.tensor(sel_work, dtype=torch.int32, device=device)
+                self.eval_offs_dev = torch.tensor(
+                    eval_offs, dtype=torch.int64, device=device
+                )
+                self.eval_offs = eval_offs
+                self.evals_dev = torch.zeros(ev_off, dtype=torch.float32, device=device)
+                self.evals_host = torch.zeros(ev_off, dtype=torch.float32, pin_memory=True)
+                nrows = len(desc_rows)
+                self.sel_table_host = torch.zeros(
+                    nrows, SEL_ROW, dtype=torch.float32, pin_memory=True
+                )
+                self.sel_table_dev = torch.zeros(
+                    nrows, SEL_ROW, dtype=torch.float32, device=device
+                )
+                self.grams_host = torch.zeros(
+                    gram_off, dtype=torch.float32, pin_memory=True
+                )
             else:
                 self.use_kernels = False
 
+    # -----------------------------------------------------------------
     def _a2d(self, grad: torch.Tensor, spec: LayerSpec) -> torch.Tensor:
         m, n = spec.meta["m"], spec.meta["n"]
         flat = grad.reshape(-1)
@@ -138,15 +155,64 @@ class BatchedSVDEncoder:
             return scratch.view(m, n)
         return flat.view(m, n)
 
+    def _sample_all(self, svals_h: dict) -> dict:
+        """Vectorized Bernoulli sampling over every layer (reference
+        _sample_svd semantics, svd.py:49-67).  svals_h: layer -> fp32/fp64
+        singular values (descending).  Returns layer -> (idx, probs)."""
+        samples = {}
+        if not self.codec.random_sample:
+            for i, spec in enumerate(self.specs):
+                r_max = spec.meta["r_max"]
+                r = min(self.codec.rank, r_max) if self.codec.rank > 0 else r_max
+                samples[i] = (torch.arange(r), None)
+            return samples
+        rank = self.codec.rank
+        probs_list = []
+        for i in range(len(self.specs)):
+            s = svals_h[i].float()
+            if s.numel() == 0 or float(s[0]) < 1e-6:
+                probs_list.append(torch.zeros(0))
+                samples[i] = (torch.tensor([0]), torch.tensor([1.0]))
+                continue
+            p = (s / s[0]) if rank == 0 else (rank * s / s.sum())
+            probs_list.append(p.clamp(max=1.0))
+        cat = torch.cat(probs_list) if probs_list else torch.zeros(0)
+        draws = torch.rand(cat.shape, generator=self.codec.generator) < cat
+        off = 0
+        for i in range(len(self.specs)):
+            p = probs_list[i]
+            k = p.numel()
+            d = draws[off : off + k]
+            off += k
+            if i in samples:
+                continue
+            idx = d.nonzero(as_tuple=False).flatten()
+            if idx.numel() == 0:  # rare: redraw this layer alone
+                idx, pr = sample_svd(
+                    svals_h[i].float(), rank=rank, generator=self.codec.generator
+                )
+                samples[i] = (idx, pr)
+            else:
+                samples[i] = (idx, p[idx])
+        # wire-budget cap: keep the highest-probability atoms
+        for i, spec in enumerate(self.specs):
+            idx, pr = samples[i]
+            r_max = spec.meta["r_max"]
+            if idx.numel() > r_max:
+                self.codec.overflow_count += 1
+                samples[i] = (idx[:r_max], None if pr is None else pr[:r_max])
+        return samples
+
     @torch.no_grad()
     def encode_all(
         self,
         grads: List[torch.Tensor],
         wire: torch.Tensor,
-        flat_grad: torch.Tensor | None = None,
+        flat_grad: Optional[torch.Tensor] = None,
     ) -> int:
         use_kernels = self.use_kernels and flat_grad is not None
         kernel_set = self.kernel_set if use_kernels else set()
+        specs = self.specs
         marks = [time.perf_counter()] if _TRACE else None
 
         def mark(label):
@@ -156,21 +222,21 @@ class BatchedSVDEncoder:
                 marks.append(time.perf_counter())
                 print(f"[enc] {label}: {1e3*(marks[-1]-marks[-2]):.2f} ms", flush=True)
 
-        specs = self.specs
-        a2ds = [self._a2d(g, s) for g, s in zip(grads, specs)]
+        host_layers = [i for i in range(len(specs)) if i not in kernel_set]
+        a2ds = {i: self._a2d(grads[i], specs[i]) for i in host_layers}
 
-        # ---- phase A: Grams on device (async) --------------------------
+        # ---- phase A: Grams (batched kernel + rocBLAS leftovers) -------
         if use_kernels:
             from ..ops import ext
 
+            e = ext()
             self.grams.zero_()
-            ext().batched_gram(
+            e.batched_gram(
                 flat_grad, self.grams, self.desc, self.gram_work,
                 self.gram_work.shape[0],
             )
-        for i, (a, s) in enumerate(zip(a2ds, specs)):
-            if i in kernel_set:
-                continue
+        for i in host_layers:
+            a = a2ds[i]
             sm = self.small[i]
             gv = self.grams[self.gram_offsets[i] : self.gram_offsets[i] + sm * sm].view(
                 sm, sm
@@ -181,38 +247,40 @@ class BatchedSVDEncoder:
                 torch.mm(a, a.t(), out=gv)
         mark("A grams")
 
-        # ---- phase B: eigensolves + ONE sync + sampling ----------------
-        by_dim = defaultdict(list)
-        for i in range(len(specs)):
-            by_dim[self.small[i]].append(i)
-        evecs_h, svals_h = {}, {}
-        if self.eigh_on_device:
-            # batched hipSOLVER eigh per distinct small-dim, ONE D2H after
-            results = []
-            for sm, idxs in by_dim.items():
-                gs = torch.stack(
-                    [
-                        self.grams[
-                            self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
-                        ].view(sm, sm)
-                        for i in idxs
-                    ]
-                )
-                gs = 0.5 * (gs + gs.transpose(1, 2))
-                evals, evecs = torch.linalg.eigh(gs)
-                results.append((idxs, evals, evecs))
-            for idxs, evals, evecs in results:
-                evals_h = evals.to("cpu", torch.float64)
-                evecs_hh = evecs.to("cpu", torch.float64)
-                evals_h = evals_h.flip(1).clamp(min=0.0)
-                evecs_hh = evecs_hh.flip(2)
-                for j, i in enumerate(idxs):
-                    svals_h[i] = evals_h[j].sqrt()
-                    evecs_h[i] = evecs_hh[j]
+        # ---- eigensolves ----------------------------------------------
+        if use_kernels:
+            e.jacobi_eigh(
+                self.grams, self.evals_dev, self.desc, self.eval_offs_dev,
+                len(self.kernel_rows),
+            )
+        svals_h, evecs_h = {}, {}
+        grams_host = None
+        if self.device.type == "cuda":
+            if use_kernels:
+                self.evals_host.copy_(self.evals_dev, non_blocking=True)
+                if host_layers:
+                    self.grams_host.copy_(self.grams, non_blocking=True)
+                    grams_host = self.grams_host
+                torch.cuda.synchronize()  # the ONE sync
+            elif host_layers:
+                grams_host = self.grams.to("cpu")  # synchronous copy
         else:
-            grams_host = self.grams.to("cpu", non_blocking=False)
-            mark("B d2h")
-            for sm, idxs in by_dim.items():
+            grams_host = self.grams
+        mark("B d2h")
+
+        if use_kernels:
+            for row, i in enumerate(self.kernel_rows):
+                sm = self.small[i]
+                o = self.eval_offs[row]
+                svals_h[i] = self.evals_host[o : o + sm].clamp(min=0.0).sqrt()
+        if host_layers:
+            by_dim = defaultdict(list)
+            for i in host_layers:
+                by_dim[self.small[i]].append(i)
+
+            def _solve_group(item):
+                sm, idxs = item
+                dt = torch.float32 if sm >= 96 else torch.float64
                 gs = torch.stack(
                     [
                         grams_host[
@@ -220,74 +288,64 @@ class BatchedSVDEncoder:
                         ].view(sm, sm)
                         for i in idxs
                     ]
-                ).to(torch.float64)
-                gs = 0.5 * (gs + gs.transpose(1, 2))  # symmetrize fp32 roundoff
-                evals, evecs = torch.linalg.eigh(gs)  # ascending
-                evals = evals.flip(1).clamp(min=0.0)
-                evecs = evecs.flip(2)
+                ).to(dt)
+                gs = 0.5 * (gs + gs.transpose(1, 2))
+                evals, evecs = torch.linalg.eigh(gs)
+                evals = evals.flip(1).clamp(min=0.0).to(torch.float64)
+                evecs = evecs.flip(2).to(torch.float64)
+                return idxs, evals, evecs
+
+            items = list(by_dim.items())
+            if len(items) > 1 and self.device.type == "cuda":
+                results = list(self._pool.map(_solve_group, items))
+            else:
+                results = [_solve_group(it) for it in items]
+            for idxs, evals, evecs in results:
                 for j, i in enumerate(idxs):
                     svals_h[i] = evals[j].sqrt()
                     evecs_h[i] = evecs[j]
         mark("B eigh")
 
-        # vectorized importance sampling: ONE rand over every layer's probs
-        samples = {}
-        if self.codec.random_sample:
-            rank = self.codec.rank
-            probs_list = []
-            for i in range(len(specs)):
-                s = svals_h[i].float()
-                if s.numel() == 0 or float(s[0]) < 1e-6:
-                    probs_list.append(torch.zeros(0))
-                    samples[i] = (torch.tensor([0]), torch.tensor([1.0]))
-                    continue
-                p = (s / s[0]) if rank == 0 else (rank * s / s.sum())
-                probs_list.append(p.clamp(max=1.0))
-            cat = torch.cat(probs_list) if probs_list else torch.zeros(0)
-            draws = torch.rand(cat.shape, generator=self.codec.generator) < cat
-            off = 0
-            for i in range(len(specs)):
-                if i in samples:
-                    continue
-                p = probs_list[i]
-                k = p.numel()
-                d = draws[off : off + k]
-                off += k
-                idx = d.nonzero(as_tuple=False).flatten()
-                if idx.numel() == 0:  # rare: redraw this layer alone
-                    idx, pr = sample_svd(
-                        svals_h[i].float(), rank=rank, generator=self.codec.generator
-                    )
-                    samples[i] = (idx, pr)
-                else:
-                    samples[i] = (idx, p[idx])
-
+        samples = self._sample_all(svals_h)
         used = 0
-        plans = []
-        stage = self.stage_host
         for i, spec in enumerate(specs):
+            idx, _ = samples[i]
+            used += 1 + idx.numel() * (spec.meta["m"] + spec.meta["n"] + 1)
+
+        # kernel layers: fill the selection table
+        if use_kernels:
+            st = self.sel_table_host
+            st.zero_()
+            for row, i in enumerate(self.kernel_rows):
+                idx, pr = samples[i]
+                r_hat = idx.numel()
+                st[row, 0] = float(r_hat)
+                st[row, 1 : 1 + r_hat] = idx.float()
+                if pr is None:
+                    st[row, 17 : 17 + r_hat] = 1.0
+                else:
+                    st[row, 17 : 17 + r_hat] = pr.float()
+            self.sel_table_dev.copy_(self.sel_table_host, non_blocking=True)
+
+        # host layers: build staged factors
+        stage = self.stage_host
+        host_plans = []
+        for i in host_layers:
+            spec = specs[i]
             sm = self.small[i]
             r_max = spec.meta["r_max"]
             s64 = svals_h[i]
             v64 = evecs_h[i]
-            if self.codec.random_sample:
-                idx, probs = samples[i]
-                if idx.numel() > r_max:
-                    self.codec.overflow_count += 1
-                    idx, probs = idx[:r_max], probs[:r_max]
-                s_sel = s64[idx]
-                s_wire = (s_sel / probs.to(torch.float64)).float()
-            else:
-                r = min(self.codec.rank, r_max) if self.codec.rank > 0 else r_max
-                idx = torch.arange(r)
-                s_sel = s64[idx]
+            idx, probs = samples[i]
+            s_sel = s64[idx]
+            if probs is None:
                 s_wire = s_sel.float()
+            else:
+                s_wire = (s_sel / probs.to(torch.float64)).float()
             r_hat = idx.numel()
-            fac = v64[:, idx]  # (sm, r_hat) eigenvectors of the small side
-            inv_s = torch.where(
-                s_sel > 1e-12, 1.0 / s_sel, torch.zeros_like(s_sel)
-            )
-            sel_scaled = (fac * inv_s.unsqueeze(0)).float()  # A @ this -> tall factor
+            fac = v64[:, idx]
+            inv_s = torch.where(s_sel > 1e-12, 1.0 / s_sel, torch.zeros_like(s_sel))
+            sel_scaled = (fac * inv_s.unsqueeze(0)).float()
             so = self.stage_offsets[i]
             stage[so] = float(r_hat)
             stage[so + 1 : so + 1 + r_hat] = s_wire
@@ -295,29 +353,30 @@ class BatchedSVDEncoder:
             stage[f_off : f_off + r_hat * sm] = fac.t().reshape(-1).float()
             sc_off = so + 1 + r_max * (1 + sm)
             stage[sc_off : sc_off + sm * r_hat] = sel_scaled.reshape(-1)
-            plans.append((i, r_hat))
-            used += 1 + r_hat * (spec.meta["m"] + spec.meta["n"] + 1)
+            host_plans.append((i, r_hat))
+            seg = 1 + r_max * (1 + 2 * sm)
+            self.stage_dev[so : so + seg].copy_(
+                stage[so : so + seg], non_blocking=True
+            )
         mark("B sample+stage")
 
-        # ---- phase C: one H2D + one batched kernel (+ rocBLAS leftovers)
-        self.stage_dev.copy_(self.stage_host, non_blocking=True)
-        sd = self.stage_dev
+        # ---- phase C: device stage build + batched sel + rocBLAS -------
         if use_kernels:
-            from ..ops import ext
-
-            ext().batched_sel(
-                flat_grad, wire, sd, self.desc, self.sel_work,
+            e.build_stage(
+                self.grams, self.evals_dev, self.sel_table_dev, self.stage_dev,
+                self.desc, self.eval_offs_dev, len(self.kernel_rows),
+            )
+            e.batched_sel(
+                flat_grad, wire, self.stage_dev, self.desc, self.sel_work,
                 self.sel_work.shape[0],
             )
-        for i, r_hat in plans:
-            if i in kernel_set:
-                continue
+        sd = self.stage_dev
+        for i, r_hat in host_plans:
             spec = specs[i]
             m, n, r_max = spec.meta["m"], spec.meta["n"], spec.meta["r_max"]
-            sm, tall = self.small[i], self.tall[i]
+            sm = self.small[i]
             wo = spec.wire_offset
             so = self.stage_offsets[i]
-            # header + s
             wire[wo : wo + 1].copy_(sd[so : so + 1])
             wire[wo + 1 + r_max * m : wo + 1 + r_max * m + r_hat].copy_(
                 sd[so + 1 : so + 1 + r_hat]
@@ -330,16 +389,14 @@ class BatchedSVDEncoder:
             ].view(sm, r_hat)
             a = a2ds[i]
             if self.m_is_tall[i]:
-                # uT (r_hat, m) = sel^T @ A^T ; vT (r_hat, n) = facT
                 u_out = wire[wo + 1 : wo + 1 + r_hat * m].view(r_hat, m)
                 torch.mm(sel.t(), a.t(), out=u_out)
                 v_off = wo + 1 + r_max * (m + 1)
                 wire[v_off : v_off + r_hat * n].view(r_hat, n).copy_(facT)
             else:
-                # uT (r_hat, m) = facT ; vT (r_hat, n) = sel^T @ A
                 wire[wo + 1 : wo + 1 + r_hat * m].view(r_hat, m).copy_(facT)
                 v_off = wo + 1 + r_max * (m + 1)
                 v_out = wire[v_off : v_off + r_hat * n].view(r_hat, n)
                 torch.mm(sel.t(), a, out=v_out)
-        mark("C h2d+gemms")
+        mark("C stage+sel")
         return used

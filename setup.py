@@ -22,6 +22,7 @@ ext = CUDAExtension(
         "atomo_amd/ops/csrc/bindings.cpp",
         "atomo_amd/ops/csrc/atomo_kernels.hip",
         "atomo_amd/ops/csrc/svd_batched.hip",
+        "atomo_amd/ops/csrc/jacobi_eigh.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -41,14 +41,23 @@ def test_gpu_loop(code, dev):
 
     trainer = _trainer(code, dev)
     train, _ = make_loaders("cifar10", 64, 64, dev, seed=3)
+    it = iter_cycle(train)
     losses = []
-    for i, (x, y) in enumerate(train):
+    for _ in range(30):
+        x, y = next(it)
         losses.append(trainer.train_step(x, y))
-        if i >= 11:
-            break
     assert all(not math.isnan(l) for l in losses)
-    assert losses[-1] < losses[0], losses
+    # compressed (sampled) gradients are noisy: compare window means
+    first = sum(losses[:5]) / 5
+    last = sum(losses[-5:]) / 5
+    assert last < first, (first, last, losses)
     assert torch.isfinite(trainer.flat).all()
+
+
+def iter_cycle(loader):
+    while True:
+        for b in loader:
+            yield b
 
 
 def test_gpu_raw_equals_plain_sgd(dev):
