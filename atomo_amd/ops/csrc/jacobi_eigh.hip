@@ -26,10 +26,11 @@
 #define JMAX 64
 #define JSTRIDE 65
 #define SWEEPS 10
+#define JTHREADS 256
 
 namespace {
 
-__global__ void __launch_bounds__(64) jacobi_eigh_kernel(
+__global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
     float* __restrict__ grams, float* __restrict__ evals,
     const int64_t* __restrict__ desc, const int64_t* __restrict__ eval_offs,
     int n_layers) {
@@ -38,6 +39,8 @@ __global__ void __launch_bounds__(64) jacobi_eigh_kernel(
   __shared__ float cs[JMAX / 2], sn[JMAX / 2];
   __shared__ int pp[JMAX / 2], qq[JMAX / 2];
   __shared__ int order[JMAX];
+  __shared__ float offsq[JTHREADS / 64];
+  __shared__ int done_s;
 
   const int layer = blockIdx.x;
   if (layer >= n_layers) return;
@@ -51,15 +54,26 @@ __global__ void __launch_bounds__(64) jacobi_eigh_kernel(
   const int N = (sm + 1) & ~1;  // even-padded
 
   // load G, init V = I (pad rows/cols zero)
-  for (int i = tid; i < N * N; i += 64) {
+  float fro2 = 0.f;
+  for (int i = tid; i < N * N; i += JTHREADS) {
     const int r = i / N, c = i % N;
-    G[r * JSTRIDE + c] = (r < sm && c < sm) ? Gg[r * sm + c] : 0.f;
+    const float g = (r < sm && c < sm) ? Gg[r * sm + c] : 0.f;
+    G[r * JSTRIDE + c] = g;
     V[r * JSTRIDE + c] = (r == c) ? 1.f : 0.f;
+    fro2 += g * g;
   }
+  // block-reduce the Frobenius norm (convergence scale)
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) fro2 += __shfl_xor(fro2, off, 64);
+  if ((tid & 63) == 0) offsq[tid >> 6] = fro2;
+  if (tid == 0) done_s = 0;
   __syncthreads();
+  float fro_all = 0.f;
+  for (int w = 0; w < JTHREADS / 64; ++w) fro_all += offsq[w];
+  const float tol2 = fro_all * 1e-13f;
 
   const int np = N / 2;
-  for (int sweep = 0; sweep < SWEEPS; ++sweep) {
+  for (int sweep = 0; sweep < SWEEPS && !done_s; ++sweep) {
     for (int round = 0; round < N - 1; ++round) {
       // tournament pairing: slot 0 fixed, others rotate
       if (tid < np) {
@@ -87,7 +101,7 @@ __global__ void __launch_bounds__(64) jacobi_eigh_kernel(
       }
       __syncthreads();
       // phase 1: rows p,q <- J^T G
-      for (int i = tid; i < np * N; i += 64) {
+      for (int i = tid; i < np * N; i += JTHREADS) {
         const int pr = i / N, k = i % N;
         const int p = pp[pr], q = qq[pr];
         const float c = cs[pr], s = sn[pr];
@@ -97,7 +111,7 @@ __global__ void __launch_bounds__(64) jacobi_eigh_kernel(
       }
       __syncthreads();
       // phase 2: cols p,q <- G J ; V <- V J
-      for (int i = tid; i < np * N; i += 64) {
+      for (int i = tid; i < np * N; i += JTHREADS) {
         const int pr = i / N, k = i % N;
         const int p = pp[pr], q = qq[pr];
         const float c = cs[pr], s = sn[pr];
@@ -110,6 +124,25 @@ __global__ void __launch_bounds__(64) jacobi_eigh_kernel(
       }
       __syncthreads();
     }
+    // convergence: off-diagonal Frobenius^2 below tolerance -> stop
+    float off2 = 0.f;
+    for (int i = tid; i < N * N; i += JTHREADS) {
+      const int r = i / N, c = i % N;
+      if (r != c) {
+        const float g = G[r * JSTRIDE + c];
+        off2 += g * g;
+      }
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) off2 += __shfl_xor(off2, off, 64);
+    if ((tid & 63) == 0) offsq[tid >> 6] = off2;
+    __syncthreads();
+    if (tid == 0) {
+      float t = 0.f;
+      for (int w = 0; w < JTHREADS / 64; ++w) t += offsq[w];
+      if (t <= tol2) done_s = 1;
+    }
+    __syncthreads();
   }
 
   // sort eigenvalues descending (insertion sort by one lane; sm <= 64)
@@ -129,12 +162,12 @@ __global__ void __launch_bounds__(64) jacobi_eigh_kernel(
   __syncthreads();
   // write sorted evals and evecs (evecs overwrite the gram slot, row-major
   // [k][j] = V[k][order[j]])
-  for (int j = tid; j < sm; j += 64) {
+  for (int j = tid; j < sm; j += JTHREADS) {
     const float lam = G[order[j] * JSTRIDE + order[j]];
     ev[j] = lam > 0.f ? lam : 0.f;
   }
   __syncthreads();
-  for (int i = tid; i < sm * sm; i += 64) {
+  for (int i = tid; i < sm * sm; i += JTHREADS) {
     const int k = i / sm, j = i % sm;
     Gg[i] = V[k * JSTRIDE + order[j]];
   }
@@ -195,7 +228,7 @@ extern "C" {
 void atomo_jacobi_eigh_launch(float* grams, float* evals, const int64_t* desc,
                               const int64_t* eval_offs, int n_layers,
                               hipStream_t stream) {
-  hipLaunchKernelGGL(jacobi_eigh_kernel, dim3(n_layers), dim3(64), 0, stream,
+  hipLaunchKernelGGL(jacobi_eigh_kernel, dim3(n_layers), dim3(JTHREADS), 0, stream,
                      grams, evals, desc, eval_offs, n_layers);
 }
 

@@ -248,59 +248,76 @@ class BatchedSVDEncoder:
         mark("A grams")
 
         # ---- eigensolves ----------------------------------------------
-        if use_kernels:
-            e.jacobi_eigh(
-                self.grams, self.evals_dev, self.desc, self.eval_offs_dev,
-                len(self.kernel_rows),
-            )
+        # Order of operations overlaps the host LAPACK eighs (big folds)
+        # with the on-device Jacobi (small folds): the Gram D2H is queued
+        # BEFORE the Jacobi launch, an event gates the host solves, and the
+        # full sync only waits for the Jacobi + evals copy.
         svals_h, evecs_h = {}, {}
         grams_host = None
+        gram_event = None
         if self.device.type == "cuda":
             if use_kernels:
-                self.evals_host.copy_(self.evals_dev, non_blocking=True)
                 if host_layers:
                     self.grams_host.copy_(self.grams, non_blocking=True)
                     grams_host = self.grams_host
-                torch.cuda.synchronize()  # the ONE sync
+                    gram_event = torch.cuda.Event()
+                    gram_event.record()
+                e.jacobi_eigh(
+                    self.grams, self.evals_dev, self.desc, self.eval_offs_dev,
+                    len(self.kernel_rows),
+                )
+                self.evals_host.copy_(self.evals_dev, non_blocking=True)
             elif host_layers:
                 grams_host = self.grams.to("cpu")  # synchronous copy
         else:
             grams_host = self.grams
-        mark("B d2h")
 
+        def _solve_group(item):
+            sm, idxs = item
+            dt = torch.float32 if sm >= 96 else torch.float64
+            gs = torch.stack(
+                [
+                    grams_host[
+                        self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
+                    ].view(sm, sm)
+                    for i in idxs
+                ]
+            ).to(dt)
+            gs = 0.5 * (gs + gs.transpose(1, 2))
+            evals, evecs = torch.linalg.eigh(gs)
+            evals = evals.flip(1).clamp(min=0.0).to(torch.float64)
+            evecs = evecs.flip(2).to(torch.float64)
+            return idxs, evals, evecs
+
+        futures = []
+        if host_layers:
+            by_dim = defaultdict(list)
+            for i in host_layers:
+                by_dim[self.small[i]].append(i)
+            items = list(by_dim.items())
+            if self.device.type == "cuda":
+                if gram_event is not None:
+                    gram_event.synchronize()
+                futures = [self._pool.submit(_solve_group, it) for it in items]
+            else:
+                futures = None
+                for it in items:
+                    idxs, evals, evecs = _solve_group(it)
+                    for j, i in enumerate(idxs):
+                        svals_h[i] = evals[j].sqrt()
+                        evecs_h[i] = evecs[j]
+
+        if self.device.type == "cuda" and use_kernels:
+            torch.cuda.synchronize()  # waits for Jacobi + evals D2H
+        mark("B d2h")
         if use_kernels:
             for row, i in enumerate(self.kernel_rows):
                 sm = self.small[i]
                 o = self.eval_offs[row]
                 svals_h[i] = self.evals_host[o : o + sm].clamp(min=0.0).sqrt()
-        if host_layers:
-            by_dim = defaultdict(list)
-            for i in host_layers:
-                by_dim[self.small[i]].append(i)
-
-            def _solve_group(item):
-                sm, idxs = item
-                dt = torch.float32 if sm >= 96 else torch.float64
-                gs = torch.stack(
-                    [
-                        grams_host[
-                            self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
-                        ].view(sm, sm)
-                        for i in idxs
-                    ]
-                ).to(dt)
-                gs = 0.5 * (gs + gs.transpose(1, 2))
-                evals, evecs = torch.linalg.eigh(gs)
-                evals = evals.flip(1).clamp(min=0.0).to(torch.float64)
-                evecs = evecs.flip(2).to(torch.float64)
-                return idxs, evals, evecs
-
-            items = list(by_dim.items())
-            if len(items) > 1 and self.device.type == "cuda":
-                results = list(self._pool.map(_solve_group, items))
-            else:
-                results = [_solve_group(it) for it in items]
-            for idxs, evals, evecs in results:
+        if futures:
+            for f in futures:
+                idxs, evals, evecs = f.result()
                 for j, i in enumerate(idxs):
                     svals_h[i] = evals[j].sqrt()
                     evecs_h[i] = evecs[j]
