@@ -65,6 +65,8 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
                    choices=["auto", "torch", "gram"])
     p.add_argument("--overlap", action="store_true", default=False,
                    help="encode layers on a side stream as backward produces them")
+    p.add_argument("--graph", action="store_true", default=False,
+                   help="capture forward/backward in a hipGraph and replay")
     return parser
 
 
@@ -111,6 +113,7 @@ class RunConfig:
             seed=a.seed,
             checkpoint_freq=a.checkpoint_freq,
             train_dir=a.train_dir,
+            use_graph=a.graph,
         )
 
 

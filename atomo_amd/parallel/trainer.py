@@ -58,6 +58,7 @@ class PSTrainer:
         seed: Optional[int] = None,
         checkpoint_freq: int = 0,
         train_dir: str = "output/models/",
+        use_graph: bool = False,
     ):
         self.comm = comm
         self.device = device or comm.device
@@ -126,6 +127,15 @@ class PSTrainer:
         self.train_dir = train_dir
         self.timers = PhaseTimers()
         self.last_loss = float("nan")
+        # hipGraph capture of forward+backward: the per-step launch storm of
+        # a convnet (hundreds of kernels) replays as one graph.  Input/label
+        # buffers are static; flat_grad.zero_() is captured inside, so
+        # autograd accumulation stays correct.
+        self.use_graph = bool(use_graph) and self.device.type == "cuda"
+        self._graph = None
+        self._static_x = None
+        self._static_y = None
+        self._static_loss = None
 
         # make every rank start from rank-0's init
         self.comm.broadcast(self.flat, src=0)
@@ -139,11 +149,14 @@ class PSTrainer:
         if self.is_worker:
             with t.phase("comp"):
                 self.model.train()
-                self.flat_grad.zero_()
-                out = self.model(x)
-                loss = self.loss_fn(out, y)
-                loss.backward()
-                self.last_loss = float(loss.detach())
+                if self.use_graph:
+                    self._fwd_bwd_graphed(x, y)
+                else:
+                    self.flat_grad.zero_()
+                    out = self.model(x)
+                    loss = self.loss_fn(out, y)
+                    loss.backward()
+                    self.last_loss = float(loss.detach())
             with t.phase("encode"):
                 if self.wc.reducible:
                     used = self.wc.total_words  # wire aliases flat_grad
@@ -186,6 +199,44 @@ class PSTrainer:
         ):
             self.save_checkpoint()
         return self.last_loss
+
+    def _fwd_bwd_graphed(self, x: torch.Tensor, y: torch.Tensor) -> None:
+        if self._graph is None:
+            try:
+                self._static_x = x.clone()
+                self._static_y = y.clone()
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    for _ in range(3):  # warmup: MIOpen algo selection
+                        self.flat_grad.zero_()
+                        loss = self.loss_fn(
+                            self.model(self._static_x), self._static_y
+                        )
+                        loss.backward()
+                torch.cuda.current_stream().wait_stream(side)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self.flat_grad.zero_()
+                    self._static_loss = self.loss_fn(
+                        self.model(self._static_x), self._static_y
+                    )
+                    self._static_loss.backward()
+                self._graph = g
+            except Exception as exc:  # capture-unsafe model: fall back
+                print(f"[atomo] hipGraph capture failed ({exc}); eager fallback",
+                      flush=True)
+                self.use_graph = False
+                self._graph = None
+                self.flat_grad.zero_()
+                loss = self.loss_fn(self.model(x), y)
+                loss.backward()
+                self.last_loss = float(loss.detach())
+                return
+        self._static_x.copy_(x)
+        self._static_y.copy_(y)
+        self._graph.replay()
+        self.last_loss = float(self._static_loss)
 
     def _apply(self, grad_flat: torch.Tensor) -> None:
         scale = 1.0 / max(1, self.num_workers)

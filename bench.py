@@ -36,6 +36,8 @@ def main():
     p.add_argument("--cpu", action="store_true", default=False)
     p.add_argument("--phase-log", action="store_true", default=False,
                    help="print per-phase timer breakdown to stderr")
+    p.add_argument("--no-graph", action="store_true", default=False,
+                   help="disable hipGraph capture of forward/backward")
     a = p.parse_args()
 
     from atomo_amd.codings import make_codec
@@ -67,6 +69,7 @@ def main():
         dedicated_ps=a.dedicated_ps,
         seed=42,
         device=device,
+        use_graph=not a.no_graph,
     )
     train, _ = make_loaders(a.dataset, a.batch_size, a.batch_size, device,
                             seed=123 + comm.rank)
