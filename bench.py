@@ -36,8 +36,10 @@ def main():
     p.add_argument("--cpu", action="store_true", default=False)
     p.add_argument("--phase-log", action="store_true", default=False,
                    help="print per-phase timer breakdown to stderr")
-    p.add_argument("--no-graph", action="store_true", default=False,
-                   help="disable hipGraph capture of forward/backward")
+    p.add_argument("--graph", action="store_true", default=False,
+                   help="capture forward/backward in a hipGraph and replay")
+    p.add_argument("--channels-last", action="store_true", default=False,
+                   help="NHWC memory format for convs")
     a = p.parse_args()
 
     from atomo_amd.codings import make_codec
@@ -69,10 +71,14 @@ def main():
         dedicated_ps=a.dedicated_ps,
         seed=42,
         device=device,
-        use_graph=not a.no_graph,
+        use_graph=a.graph,
     )
+    if a.channels_last:
+        trainer.model.to(memory_format=torch.channels_last)
     train, _ = make_loaders(a.dataset, a.batch_size, a.batch_size, device,
                             seed=123 + comm.rank)
+    if a.channels_last:
+        train.x = train.x.contiguous(memory_format=torch.channels_last)
 
     def sync():
         comm.barrier()
