@@ -21,10 +21,13 @@ void atomo_fused_sgd_launch(float*, const float*, float*, int64_t, float,
 void atomo_batched_gram_launch(const float*, float*, const int64_t*,
                                const int32_t*, int, hipStream_t);
 void atomo_batched_sel_launch(const float*, float*, const float*,
-                              const int64_t*, const int32_t*, int,
+                              const int64_t*, const int32_t*, int, int,
                               hipStream_t);
 void atomo_jacobi_eigh_launch(float*, float*, const int64_t*, const int64_t*,
-                              int, hipStream_t);
+                              const int32_t*, int, hipStream_t);
+void atomo_jacobi_eigh_big_launch(float*, float*, float*, const int64_t*,
+                                  const int64_t*, const int32_t*,
+                                  const int64_t*, int, hipStream_t);
 void atomo_build_stage_launch(const float*, const float*, const float*,
                               float*, const int64_t*, const int64_t*, int,
                               hipStream_t);
@@ -127,7 +130,8 @@ void batched_gram(torch::Tensor flat, torch::Tensor grams, torch::Tensor desc,
 }
 
 void batched_sel(torch::Tensor flat, torch::Tensor wire, torch::Tensor stage,
-                 torch::Tensor desc, torch::Tensor work, int64_t n_tiles) {
+                 torch::Tensor desc, torch::Tensor work, int64_t n_tiles,
+                 int64_t sel_elems) {
   check_f32_cuda(flat, "flat");
   check_f32_cuda(wire, "wire");
   check_f32_cuda(stage, "stage");
@@ -140,11 +144,11 @@ void batched_sel(torch::Tensor flat, torch::Tensor wire, torch::Tensor stage,
   atomo_batched_sel_launch(flat.data_ptr<float>(), wire.data_ptr<float>(),
                            stage.data_ptr<float>(), desc.data_ptr<int64_t>(),
                            work.data_ptr<int32_t>(), (int)n_tiles,
-                           cur_stream());
+                           (int)sel_elems, cur_stream());
 }
 
 void jacobi_eigh(torch::Tensor grams, torch::Tensor evals, torch::Tensor desc,
-                 torch::Tensor eval_offs, int64_t n_layers) {
+                 torch::Tensor eval_offs, torch::Tensor rows, int64_t n_mats) {
   check_f32_cuda(grams, "grams");
   check_f32_cuda(evals, "evals");
   TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kInt64 &&
@@ -152,10 +156,29 @@ void jacobi_eigh(torch::Tensor grams, torch::Tensor evals, torch::Tensor desc,
               "desc must be contiguous cuda int64");
   TORCH_CHECK(eval_offs.is_cuda() && eval_offs.scalar_type() == torch::kInt64,
               "eval_offs must be cuda int64");
+  TORCH_CHECK(rows.is_cuda() && rows.scalar_type() == torch::kInt32,
+              "rows must be cuda int32");
+  if (n_mats == 0) return;
   atomo_jacobi_eigh_launch(grams.data_ptr<float>(), evals.data_ptr<float>(),
                            desc.data_ptr<int64_t>(),
-                           eval_offs.data_ptr<int64_t>(), (int)n_layers,
+                           eval_offs.data_ptr<int64_t>(),
+                           rows.data_ptr<int32_t>(), (int)n_mats,
                            cur_stream());
+}
+
+void jacobi_eigh_big(torch::Tensor grams, torch::Tensor vbuf,
+                     torch::Tensor evals, torch::Tensor desc,
+                     torch::Tensor eval_offs, torch::Tensor rows,
+                     torch::Tensor v_offs, int64_t n_mats) {
+  check_f32_cuda(grams, "grams");
+  check_f32_cuda(vbuf, "vbuf");
+  check_f32_cuda(evals, "evals");
+  if (n_mats == 0) return;
+  atomo_jacobi_eigh_big_launch(
+      grams.data_ptr<float>(), vbuf.data_ptr<float>(),
+      evals.data_ptr<float>(), desc.data_ptr<int64_t>(),
+      eval_offs.data_ptr<int64_t>(), rows.data_ptr<int32_t>(),
+      v_offs.data_ptr<int64_t>(), (int)n_mats, cur_stream());
 }
 
 void build_stage(torch::Tensor evecs, torch::Tensor evals,
@@ -177,7 +200,9 @@ void build_stage(torch::Tensor evecs, torch::Tensor evals,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("jacobi_eigh", &jacobi_eigh,
-        "batched parallel-Jacobi symmetric eigensolver (sm <= 64)");
+        "batched parallel-Jacobi symmetric eigensolver (sm <= 64, LDS)");
+  m.def("jacobi_eigh_big", &jacobi_eigh_big,
+        "batched parallel-Jacobi eigensolver (64 < sm <= 512, L2-resident)");
   m.def("build_stage", &build_stage,
         "gather sampled atoms into the staged wire factors");
   m.def("batched_gram", &batched_gram,

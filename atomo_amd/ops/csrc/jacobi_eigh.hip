@@ -1,19 +1,24 @@
-// Batched symmetric eigensolver (cyclic parallel Jacobi) + selection-stage
+// Batched symmetric eigensolvers (cyclic parallel Jacobi) + selection-stage
 // builder for MI355X (gfx950).
 //
-// Solves the per-layer Gram matrices G (sm x sm, sm <= 64) ON DEVICE so the
-// SVD-encode path never ships Grams to the host (reference equivalent:
-// numpy LA.svd per layer, codings/svd.py:95).  One wave64 workgroup per
-// matrix; G and V live in LDS (row stride 65 to keep column walks
-// conflict-free); each Jacobi round applies all N/2 disjoint plane
-// rotations in two barrier-separated phases (rows = J^T G, then cols = .J
-// and V.J); pairs follow the round-robin tournament schedule.  Fixed sweep
-// count (machine-eps convergence for n <= 64 needs ~6; we run 10).
-// Eigenvalues are sorted descending and the eigenvectors overwrite the Gram
-// slot in-place.
+// Solves the per-layer Gram matrices ON DEVICE so the SVD-encode path never
+// ships Grams to the host (reference equivalent: numpy LA.svd per layer,
+// codings/svd.py:95).  Two variants:
 //
-// build_stage_kernel then gathers the host-sampled atom selection
-// (idx, probs per layer) into the staged wire factors:
+//   jacobi_eigh_kernel      sm <= 64: G and V in LDS (row stride 65 keeps
+//                           column walks conflict-free), 256 threads/WG.
+//   jacobi_eigh_big_kernel  64 < sm <= 512: G stays in its global Gram slot
+//                           (L2-resident: 1 MB for sm=512 fits one XCD's
+//                           4 MB L2), V in a global scratch; 512 threads/WG.
+//
+// Both: each Jacobi round applies all N/2 disjoint plane rotations in two
+// barrier-separated phases (rows = J^T G, then cols = .J and V.J); pairs
+// follow the round-robin tournament schedule; early exit when the
+// off-diagonal Frobenius norm drops below 1e-13 x ||G||_F^2.  Eigenvalues
+// sort descending; eigenvectors overwrite the Gram slot.
+//
+// build_stage_kernel gathers the host-sampled atom selection (idx, probs
+// per layer) into the staged wire factors:
 //   s_wire[r] = sqrt(eval[idx_r]) / p_r
 //   facT[r][k] = V[k][idx_r]                  (the small wire factor)
 //   sel[k][r]  = V[k][idx_r] / sqrt(eval)     (A @ sel = tall factor)
@@ -27,13 +32,32 @@
 #define JSTRIDE 65
 #define SWEEPS 10
 #define JTHREADS 256
+#define JBIG_THREADS 512
+#define R_CAP 32
+#define SEL_ROW 65  // [r_hat | idx*32 | probs*32]
 
 namespace {
 
+__device__ __forceinline__ void rot_params(float app, float aqq, float apq,
+                                           float* c, float* s) {
+  *c = 1.f;
+  *s = 0.f;
+  if (fabsf(apq) > 1e-12f) {
+    const float tau = (aqq - app) / (2.f * apq);
+    const float t =
+        (tau >= 0.f ? 1.f : -1.f) / (fabsf(tau) + sqrtf(1.f + tau * tau));
+    *c = rsqrtf(1.f + t * t);
+    *s = t * (*c);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// small variant: LDS-resident
+// ---------------------------------------------------------------------------
 __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
     float* __restrict__ grams, float* __restrict__ evals,
     const int64_t* __restrict__ desc, const int64_t* __restrict__ eval_offs,
-    int n_layers) {
+    const int32_t* __restrict__ rows_list, int n_mats) {
   __shared__ float G[JMAX * JSTRIDE];
   __shared__ float V[JMAX * JSTRIDE];
   __shared__ float cs[JMAX / 2], sn[JMAX / 2];
@@ -42,18 +66,17 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
   __shared__ float offsq[JTHREADS / 64];
   __shared__ int done_s;
 
-  const int layer = blockIdx.x;
-  if (layer >= n_layers) return;
-  const int64_t* d = desc + (int64_t)layer * GD_N;
+  if (blockIdx.x >= (unsigned)n_mats) return;
+  const int row = rows_list[blockIdx.x];
+  const int64_t* d = desc + (int64_t)row * GD_N;
   const int m = (int)d[1], n = (int)d[2];
   const bool is_tall = d[3] != 0;
   const int sm = is_tall ? n : m;
   float* Gg = grams + d[4];
-  float* ev = evals + eval_offs[layer];
+  float* ev = evals + eval_offs[row];
   const int tid = threadIdx.x;
   const int N = (sm + 1) & ~1;  // even-padded
 
-  // load G, init V = I (pad rows/cols zero)
   float fro2 = 0.f;
   for (int i = tid; i < N * N; i += JTHREADS) {
     const int r = i / N, c = i % N;
@@ -62,7 +85,6 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
     V[r * JSTRIDE + c] = (r == c) ? 1.f : 0.f;
     fro2 += g * g;
   }
-  // block-reduce the Frobenius norm (convergence scale)
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) fro2 += __shfl_xor(fro2, off, 64);
   if ((tid & 63) == 0) offsq[tid >> 6] = fro2;
@@ -75,32 +97,19 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
   const int np = N / 2;
   for (int sweep = 0; sweep < SWEEPS && !done_s; ++sweep) {
     for (int round = 0; round < N - 1; ++round) {
-      // tournament pairing: slot 0 fixed, others rotate
       if (tid < np) {
         auto player = [&](int slot) {
           return slot == 0 ? 0 : 1 + (slot - 1 + round) % (N - 1);
         };
-        int a = player(tid);
-        int b = player(N - 1 - tid);
+        const int a = player(tid);
+        const int b = player(N - 1 - tid);
         const int p = min(a, b), q = max(a, b);
         pp[tid] = p;
         qq[tid] = q;
-        const float app = G[p * JSTRIDE + p];
-        const float aqq = G[q * JSTRIDE + q];
-        const float apq = G[p * JSTRIDE + q];
-        float c = 1.f, s = 0.f;
-        if (fabsf(apq) > 1e-12f) {
-          const float tau = (aqq - app) / (2.f * apq);
-          const float t =
-              (tau >= 0.f ? 1.f : -1.f) / (fabsf(tau) + sqrtf(1.f + tau * tau));
-          c = rsqrtf(1.f + t * t);
-          s = t * c;
-        }
-        cs[tid] = c;
-        sn[tid] = s;
+        rot_params(G[p * JSTRIDE + p], G[q * JSTRIDE + q], G[p * JSTRIDE + q],
+                   &cs[tid], &sn[tid]);
       }
       __syncthreads();
-      // phase 1: rows p,q <- J^T G
       for (int i = tid; i < np * N; i += JTHREADS) {
         const int pr = i / N, k = i % N;
         const int p = pp[pr], q = qq[pr];
@@ -110,7 +119,6 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
         G[q * JSTRIDE + k] = s * gp + c * gq;
       }
       __syncthreads();
-      // phase 2: cols p,q <- G J ; V <- V J
       for (int i = tid; i < np * N; i += JTHREADS) {
         const int pr = i / N, k = i % N;
         const int p = pp[pr], q = qq[pr];
@@ -124,7 +132,6 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
       }
       __syncthreads();
     }
-    // convergence: off-diagonal Frobenius^2 below tolerance -> stop
     float off2 = 0.f;
     for (int i = tid; i < N * N; i += JTHREADS) {
       const int r = i / N, c = i % N;
@@ -145,7 +152,6 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
     __syncthreads();
   }
 
-  // sort eigenvalues descending (insertion sort by one lane; sm <= 64)
   if (tid == 0) {
     for (int i = 0; i < sm; ++i) order[i] = i;
     for (int i = 1; i < sm; ++i) {
@@ -160,8 +166,6 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
     }
   }
   __syncthreads();
-  // write sorted evals and evecs (evecs overwrite the gram slot, row-major
-  // [k][j] = V[k][order[j]])
   for (int j = tid; j < sm; j += JTHREADS) {
     const float lam = G[order[j] * JSTRIDE + order[j]];
     ev[j] = lam > 0.f ? lam : 0.f;
@@ -174,19 +178,150 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// big variant: G in its global Gram slot, V in global scratch
+// (both L2-resident for sm <= 512)
+// ---------------------------------------------------------------------------
+#define BIGMAX 512
+
+__global__ void __launch_bounds__(JBIG_THREADS) jacobi_eigh_big_kernel(
+    float* __restrict__ grams, float* __restrict__ vbuf,
+    float* __restrict__ evals, const int64_t* __restrict__ desc,
+    const int64_t* __restrict__ eval_offs,
+    const int32_t* __restrict__ rows_list,
+    const int64_t* __restrict__ v_offs, int n_mats) {
+  __shared__ float cs[BIGMAX / 2], sn[BIGMAX / 2];
+  __shared__ short pp[BIGMAX / 2], qq[BIGMAX / 2];
+  __shared__ float diag[BIGMAX];
+  __shared__ short order[BIGMAX];
+  __shared__ float offsq[JBIG_THREADS / 64];
+  __shared__ int done_s;
+
+  if (blockIdx.x >= (unsigned)n_mats) return;
+  const int row = rows_list[blockIdx.x];
+  const int64_t* d = desc + (int64_t)row * GD_N;
+  const int m = (int)d[1], n = (int)d[2];
+  const bool is_tall = d[3] != 0;
+  const int sm = is_tall ? n : m;
+  float* G = grams + d[4];  // (sm, sm) row-major, in place
+  float* V = vbuf + v_offs[blockIdx.x];
+  float* ev = evals + eval_offs[row];
+  const int tid = threadIdx.x;
+  // sm is even for every real fold here (tall folds have even a*b or this
+  // layer went to the host path); assume N == sm.
+  const int N = sm;
+  const int np = N / 2;
+
+  float fro2 = 0.f;
+  for (int i = tid; i < N * N; i += JBIG_THREADS) {
+    const float g = G[i];
+    V[i] = (i / N == i % N) ? 1.f : 0.f;
+    fro2 += g * g;
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) fro2 += __shfl_xor(fro2, off, 64);
+  if ((tid & 63) == 0) offsq[tid >> 6] = fro2;
+  if (tid == 0) done_s = 0;
+  __syncthreads();
+  float fro_all = 0.f;
+  for (int w = 0; w < JBIG_THREADS / 64; ++w) fro_all += offsq[w];
+  const float tol2 = fro_all * 1e-13f;
+
+  for (int sweep = 0; sweep < SWEEPS && !done_s; ++sweep) {
+    for (int round = 0; round < N - 1; ++round) {
+      for (int i = tid; i < np; i += JBIG_THREADS) {
+        auto player = [&](int slot) {
+          return slot == 0 ? 0 : 1 + (slot - 1 + round) % (N - 1);
+        };
+        const int a = player(i);
+        const int b = player(N - 1 - i);
+        const int p = min(a, b), q = max(a, b);
+        pp[i] = (short)p;
+        qq[i] = (short)q;
+        rot_params(G[(int64_t)p * N + p], G[(int64_t)q * N + q],
+                   G[(int64_t)p * N + q], &cs[i], &sn[i]);
+      }
+      __syncthreads();
+      for (int i = tid; i < np * N; i += JBIG_THREADS) {
+        const int pr = i / N, k = i % N;
+        const int p = pp[pr], q = qq[pr];
+        const float c = cs[pr], s = sn[pr];
+        const float gp = G[(int64_t)p * N + k], gq = G[(int64_t)q * N + k];
+        G[(int64_t)p * N + k] = c * gp - s * gq;
+        G[(int64_t)q * N + k] = s * gp + c * gq;
+      }
+      __syncthreads();
+      for (int i = tid; i < np * N; i += JBIG_THREADS) {
+        const int pr = i / N, k = i % N;
+        const int p = pp[pr], q = qq[pr];
+        const float c = cs[pr], s = sn[pr];
+        const float gp = G[(int64_t)k * N + p], gq = G[(int64_t)k * N + q];
+        G[(int64_t)k * N + p] = c * gp - s * gq;
+        G[(int64_t)k * N + q] = s * gp + c * gq;
+        const float vp = V[(int64_t)k * N + p], vq = V[(int64_t)k * N + q];
+        V[(int64_t)k * N + p] = c * vp - s * vq;
+        V[(int64_t)k * N + q] = s * vp + c * vq;
+      }
+      __syncthreads();
+    }
+    float off2 = 0.f;
+    for (int i = tid; i < N * N; i += JBIG_THREADS) {
+      if (i / N != i % N) {
+        const float g = G[i];
+        off2 += g * g;
+      }
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) off2 += __shfl_xor(off2, off, 64);
+    if ((tid & 63) == 0) offsq[tid >> 6] = off2;
+    __syncthreads();
+    if (tid == 0) {
+      float t = 0.f;
+      for (int w = 0; w < JBIG_THREADS / 64; ++w) t += offsq[w];
+      if (t <= tol2) done_s = 1;
+    }
+    __syncthreads();
+  }
+
+  // stage the diagonal in LDS and sort
+  for (int i = tid; i < N; i += JBIG_THREADS) diag[i] = G[(int64_t)i * N + i];
+  __syncthreads();
+  if (tid == 0) {
+    for (int i = 0; i < N; ++i) order[i] = (short)i;
+    for (int i = 1; i < N; ++i) {
+      const short oi = order[i];
+      const float vi = diag[oi];
+      int j = i - 1;
+      while (j >= 0 && diag[order[j]] < vi) {
+        order[j + 1] = order[j];
+        --j;
+      }
+      order[j + 1] = oi;
+    }
+  }
+  __syncthreads();
+  for (int j = tid; j < N; j += JBIG_THREADS) {
+    const float lam = diag[order[j]];
+    ev[j] = lam > 0.f ? lam : 0.f;
+  }
+  // permute eigenvectors into the gram slot: Gg[k][j] = V[k][order[j]]
+  for (int i = tid; i < N * N; i += JBIG_THREADS) {
+    const int k = i / N, j = i % N;
+    G[i] = V[(int64_t)k * N + order[j]];
+  }
+}
+
+// ---------------------------------------------------------------------------
 // stage builder: per layer gather the sampled atoms into the wire staging
-// sel table row (fp32): [r_hat | idx 0..15 (as float) | probs 0..15]
+// sel table row (fp32): [r_hat | idx 0..31 (as float) | probs 0..31]
 // stage layout:          [r_hat | s_wire(r_max) | facT(r_max*sm) | sel(sm*r_max)]
 // ---------------------------------------------------------------------------
-#define SEL_ROW 33
-
 __global__ void __launch_bounds__(64) build_stage_kernel(
     const float* __restrict__ evecs, const float* __restrict__ evals,
     const float* __restrict__ sel_table, float* __restrict__ stage,
     const int64_t* __restrict__ desc, const int64_t* __restrict__ eval_offs,
     int n_layers) {
-  __shared__ float s_sel[16], inv_s[16];
-  __shared__ int idxs[16];
+  __shared__ float s_sel[R_CAP], inv_s[R_CAP];
+  __shared__ int idxs[R_CAP];
   const int layer = blockIdx.x;
   if (layer >= n_layers) return;
   const int64_t* d = desc + (int64_t)layer * GD_N;
@@ -203,7 +338,7 @@ __global__ void __launch_bounds__(64) build_stage_kernel(
   if (tid == 0) stage[so] = (float)r_hat;
   if (tid < r_hat) {
     const int idx = (int)row[1 + tid];
-    const float p = row[17 + tid];
+    const float p = row[1 + R_CAP + tid];
     const float s = sqrtf(ev[idx]);
     idxs[tid] = idx;
     s_sel[tid] = s / p;          // shipped singular value (unbiased rescale)
@@ -211,7 +346,7 @@ __global__ void __launch_bounds__(64) build_stage_kernel(
     stage[so + 1 + tid] = s_sel[tid];
   }
   __syncthreads();
-  float* facT = stage + so + 1 + r_max;                  // (r_hat, sm)
+  float* facT = stage + so + 1 + r_max;                     // (r_hat, sm)
   float* sel = stage + so + 1 + (int64_t)r_max * (1 + sm);  // (sm, r_hat)
   for (int i = tid; i < r_hat * sm; i += 64) {
     const int r = i / sm, k = i % sm;
@@ -226,10 +361,22 @@ __global__ void __launch_bounds__(64) build_stage_kernel(
 extern "C" {
 
 void atomo_jacobi_eigh_launch(float* grams, float* evals, const int64_t* desc,
-                              const int64_t* eval_offs, int n_layers,
+                              const int64_t* eval_offs,
+                              const int32_t* rows_list, int n_mats,
                               hipStream_t stream) {
-  hipLaunchKernelGGL(jacobi_eigh_kernel, dim3(n_layers), dim3(JTHREADS), 0, stream,
-                     grams, evals, desc, eval_offs, n_layers);
+  hipLaunchKernelGGL(jacobi_eigh_kernel, dim3(n_mats), dim3(JTHREADS), 0,
+                     stream, grams, evals, desc, eval_offs, rows_list, n_mats);
+}
+
+void atomo_jacobi_eigh_big_launch(float* grams, float* vbuf, float* evals,
+                                  const int64_t* desc,
+                                  const int64_t* eval_offs,
+                                  const int32_t* rows_list,
+                                  const int64_t* v_offs, int n_mats,
+                                  hipStream_t stream) {
+  hipLaunchKernelGGL(jacobi_eigh_big_kernel, dim3(n_mats), dim3(JBIG_THREADS),
+                     0, stream, grams, vbuf, evals, desc, eval_offs, rows_list,
+                     v_offs, n_mats);
 }
 
 void atomo_build_stage_launch(const float* evecs, const float* evals,

@@ -138,9 +138,11 @@ __device__ __forceinline__ void sel_rows(const float* __restrict__ A,
 __global__ void __launch_bounds__(256) batched_sel_kernel(
     const float* __restrict__ flat, float* __restrict__ wire,
     const float* __restrict__ stage, const int64_t* __restrict__ desc,
-    const int32_t* __restrict__ work, int n_tiles) {
-  __shared__ float sel_lds[64 * 16];  // sm x r_hat, sm <= 64, r_hat <= 16
-  __shared__ int cached_layer_s;
+    const int32_t* __restrict__ work, int n_tiles, int sel_elems) {
+  // dynamic LDS: [sel (sel_elems floats)] [cached_layer (1 int)]
+  extern __shared__ __attribute__((aligned(16))) float sel_lds[];
+  int* cached_layer_p = (int*)(sel_lds + sel_elems);
+#define cached_layer_s (*cached_layer_p)
   if (threadIdx.x == 0) cached_layer_s = -1;
   __syncthreads();
   for (int tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
@@ -186,10 +188,14 @@ __global__ void __launch_bounds__(256) batched_sel_kernel(
       sel_rows<4>(A, wire, sel_lds, t_begin, tall, m, n, is_tall, r_hat, t_dst);
     else if (r_hat <= 8)
       sel_rows<8>(A, wire, sel_lds, t_begin, tall, m, n, is_tall, r_hat, t_dst);
-    else
+    else if (r_hat <= 16)
       sel_rows<16>(A, wire, sel_lds, t_begin, tall, m, n, is_tall, r_hat,
                    t_dst);
+    else
+      sel_rows<32>(A, wire, sel_lds, t_begin, tall, m, n, is_tall, r_hat,
+                   t_dst);
   }
+#undef cached_layer_s
 }
 
 inline int grid_for_tiles(int tiles) {
@@ -212,11 +218,12 @@ void atomo_batched_gram_launch(const float* flat, float* grams,
 
 void atomo_batched_sel_launch(const float* flat, float* wire,
                               const float* stage, const int64_t* desc,
-                              const int32_t* work, int n_tiles,
+                              const int32_t* work, int n_tiles, int sel_elems,
                               hipStream_t stream) {
+  const size_t lds = (size_t)sel_elems * 4 + 16;
   hipLaunchKernelGGL(batched_sel_kernel, dim3(grid_for_tiles(n_tiles)),
-                     dim3(256), 0, stream, flat, wire, stage, desc, work,
-                     n_tiles);
+                     dim3(256), lds, stream, flat, wire, stage, desc, work,
+                     n_tiles, sel_elems);
 }
 
 }  // extern "C"

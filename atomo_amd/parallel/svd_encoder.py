@@ -4,16 +4,16 @@ The naive per-layer path costs ~60 host round trips per step (factorize,
 sample, write).  This encoder restructures the work MI355X-first; layers
 split into two classes:
 
-KERNEL layers (small-dim <= 64, even fold — every conv/BN/bias and most fc):
-  entirely on device via the hand-written gfx950 kernels
-  (ops/csrc/svd_batched.hip, ops/csrc/jacobi_eigh.hip):
-    batched_gram -> jacobi_eigh (parallel cyclic Jacobi, one wave64 per
-    matrix) -> [evals to host: the ONLY sync] -> host Bernoulli sampling
-    (vectorized across layers) -> sel_table H2D -> build_stage ->
-    batched_sel writes the wire packets.
-HOST layers (small-dim > 64: a few 1x1-conv/fc folds, odd-padded 1-D):
-  Gram via rocBLAS, fp64/fp32 LAPACK eigh on a thread pool, staged
-  selection factors H2D, rocBLAS selection GEMMs straight into the wire.
+DEVICE layers (even 2-D fold, wire budget r_max <= 32, small-dim <= 512 —
+every layer of the BASELINE model configs):
+  batched_gram (hand-written kernel, small-dim <= 64) or rocBLAS Gram
+  (bigger folds) -> jacobi_eigh / jacobi_eigh_big (parallel cyclic Jacobi:
+  LDS-resident for sm <= 64, L2-resident global for 64 < sm <= 512) ->
+  [evals to host: the ONLY sync] -> vectorized host Bernoulli sampling ->
+  sel_table H2D -> build_stage -> batched_sel writes the wire packets.
+HOST layers (odd zero-padded 1-D folds, or r_max > 32 / sm > 512):
+  rocBLAS Gram, fp64/fp32 LAPACK eigh on a thread pool, staged factors H2D,
+  rocBLAS selection GEMMs straight into the wire.
 
 Semantics identical to SVDCodec.encode_into (same wire layout, same
 sampler, same unbiasedness invariant E[sum s_i/p_i u_i v_i^T] = grad;
@@ -36,8 +36,10 @@ _TRACE = os.environ.get("ATOMO_TRACE_ENCODER", "") not in ("", "0")
 
 GRAM_CHUNK = 256
 SEL_CHUNK = 1024
-SEL_ROW = 33  # [r_hat | idx*16 | probs*16]
-R_CAP = 16
+R_CAP = 32
+SEL_ROW = 1 + 2 * R_CAP  # [r_hat | idx*32 | probs*32]
+SMALL_SM = 64
+BIG_SM = 512
 
 
 class BatchedSVDEncoder:
@@ -86,8 +88,9 @@ class BatchedSVDEncoder:
         }
 
         # ---- descriptor tables for the batched HIP kernels -------------
-        self.kernel_set = set()
+        self.kernel_set = set()  # device-path layer indices
         self.kernel_rows: List[int] = []  # desc row -> layer index
+        self.big_gram_layers: List[int] = []  # device rows needing rocBLAS gram
         self.use_kernels = False
         if device.type == "cuda" and param_offsets is not None:
             from .. import ops
@@ -95,11 +98,18 @@ class BatchedSVDEncoder:
             self.use_kernels = ops.have_ext()
         if self.use_kernels:
             desc_rows, gram_work, sel_work, eval_offs = [], [], [], []
-            ev_off = 0
+            small_rows, big_rows, big_v_offs = [], [], []
+            ev_off, v_off, sel_elems = 0, 0, 1
             for i, s in enumerate(specs):
                 m, n = s.meta["m"], s.meta["n"]
                 sm, tall = self.small[i], self.tall[i]
-                if s.meta["padded"] != s.numel or sm > 64 or s.meta["r_max"] > R_CAP:
+                device_ok = (
+                    s.meta["padded"] == s.numel
+                    and s.meta["r_max"] <= R_CAP
+                    and sm <= BIG_SM
+                    and (sm <= SMALL_SM or sm % 2 == 0)
+                )
+                if not device_ok:
                     continue
                 row = len(desc_rows)
                 desc_rows.append(
@@ -118,30 +128,36 @@ class BatchedSVDEncoder:
                 ev_off += sm
                 self.kernel_set.add(i)
                 self.kernel_rows.append(i)
-                for c in range((tall + GRAM_CHUNK - 1) // GRAM_CHUNK):
-                    gram_work.append([row, c])
+                sel_elems = max(sel_elems, sm * min(R_CAP, s.meta["r_max"]))
+                if sm <= SMALL_SM:
+                    small_rows.append(row)
+                    for c in range((tall + GRAM_CHUNK - 1) // GRAM_CHUNK):
+                        gram_work.append([row, c])
+                else:
+                    big_rows.append(row)
+                    big_v_offs.append(v_off)
+                    v_off += sm * sm
+                    self.big_gram_layers.append(i)
                 for c in range((tall + SEL_CHUNK - 1) // SEL_CHUNK):
                     sel_work.append([row, c])
             if desc_rows:
-                self.desc = torch.tensor(desc_rows, dtype=torch.int64, device=device)
-                self.gram_work = torch.tensor(gram_work, dtype=torch.int32, device=device)
-                self.sel_work = torch.tensor(sel_work, dtype=torch.int32, device=device)
-                self.eval_offs_dev = torch.tensor(
-                    eval_offs, dtype=torch.int64, device=device
-                )
+                dev = device
+                self.desc = torch.tensor(desc_rows, dtype=torch.int64, device=dev)
+                self.gram_work = torch.tensor(gram_work, dtype=torch.int32, device=dev)
+                self.sel_work = torch.tensor(sel_work, dtype=torch.int32, device=dev)
+                self.small_rows = torch.tensor(small_rows, dtype=torch.int32, device=dev)
+                self.big_rows = torch.tensor(big_rows, dtype=torch.int32, device=dev)
+                self.big_v_offs = torch.tensor(big_v_offs, dtype=torch.int64, device=dev)
+                self.vbuf = torch.zeros(max(1, v_off), dtype=torch.float32, device=dev)
+                self.eval_offs_dev = torch.tensor(eval_offs, dtype=torch.int64, device=dev)
                 self.eval_offs = eval_offs
-                self.evals_dev = torch.zeros(ev_off, dtype=torch.float32, device=device)
-                self.evals_host = torch.zeros(ev_off, dtype=torch.float32, pin_memory=True)
+                self.evals_dev = torch.zeros(max(1, ev_off), dtype=torch.float32, device=dev)
+                self.evals_host = torch.zeros(max(1, ev_off), dtype=torch.float32, pin_memory=True)
                 nrows = len(desc_rows)
-                self.sel_table_host = torch.zeros(
-                    nrows, SEL_ROW, dtype=torch.float32, pin_memory=True
-                )
-                self.sel_table_dev = torch.zeros(
-                    nrows, SEL_ROW, dtype=torch.float32, device=device
-                )
-                self.grams_host = torch.zeros(
-                    gram_off, dtype=torch.float32, pin_memory=True
-                )
+                self.sel_table_host = torch.zeros(nrows, SEL_ROW, dtype=torch.float32, pin_memory=True)
+                self.sel_table_dev = torch.zeros(nrows, SEL_ROW, dtype=torch.float32, device=dev)
+                self.grams_host = torch.zeros(gram_off, dtype=torch.float32, pin_memory=True)
+                self.sel_elems = sel_elems
             else:
                 self.use_kernels = False
 
@@ -223,7 +239,8 @@ class BatchedSVDEncoder:
                 print(f"[enc] {label}: {1e3*(marks[-1]-marks[-2]):.2f} ms", flush=True)
 
         host_layers = [i for i in range(len(specs)) if i not in kernel_set]
-        a2ds = {i: self._a2d(grads[i], specs[i]) for i in host_layers}
+        mm_layers = host_layers + (self.big_gram_layers if use_kernels else [])
+        a2ds = {i: self._a2d(grads[i], specs[i]) for i in mm_layers}
 
         # ---- phase A: Grams (batched kernel + rocBLAS leftovers) -------
         if use_kernels:
@@ -231,11 +248,12 @@ class BatchedSVDEncoder:
 
             e = ext()
             self.grams.zero_()
-            e.batched_gram(
-                flat_grad, self.grams, self.desc, self.gram_work,
-                self.gram_work.shape[0],
-            )
-        for i in host_layers:
+            if self.gram_work.shape[0]:
+                e.batched_gram(
+                    flat_grad, self.grams, self.desc, self.gram_work,
+                    self.gram_work.shape[0],
+                )
+        for i in mm_layers:
             a = a2ds[i]
             sm = self.small[i]
             gv = self.grams[self.gram_offsets[i] : self.gram_offsets[i] + sm * sm].view(
@@ -248,10 +266,9 @@ class BatchedSVDEncoder:
         mark("A grams")
 
         # ---- eigensolves ----------------------------------------------
-        # Order of operations overlaps the host LAPACK eighs (big folds)
-        # with the on-device Jacobi (small folds): the Gram D2H is queued
-        # BEFORE the Jacobi launch, an event gates the host solves, and the
-        # full sync only waits for the Jacobi + evals copy.
+        # host LAPACK (odd/oversize layers) overlaps the device Jacobi:
+        # the Gram D2H is queued BEFORE the Jacobi launches, an event gates
+        # the host solves, and the full sync waits for Jacobi + evals copy.
         svals_h, evecs_h = {}, {}
         grams_host = None
         gram_event = None
@@ -264,7 +281,12 @@ class BatchedSVDEncoder:
                     gram_event.record()
                 e.jacobi_eigh(
                     self.grams, self.evals_dev, self.desc, self.eval_offs_dev,
-                    len(self.kernel_rows),
+                    self.small_rows, self.small_rows.shape[0],
+                )
+                e.jacobi_eigh_big(
+                    self.grams, self.vbuf, self.evals_dev, self.desc,
+                    self.eval_offs_dev, self.big_rows, self.big_v_offs,
+                    self.big_rows.shape[0],
                 )
                 self.evals_host.copy_(self.evals_dev, non_blocking=True)
             elif host_layers:
@@ -329,7 +351,7 @@ class BatchedSVDEncoder:
             idx, _ = samples[i]
             used += 1 + idx.numel() * (spec.meta["m"] + spec.meta["n"] + 1)
 
-        # kernel layers: fill the selection table
+        # device layers: fill the selection table
         if use_kernels:
             st = self.sel_table_host
             st.zero_()
@@ -339,9 +361,9 @@ class BatchedSVDEncoder:
                 st[row, 0] = float(r_hat)
                 st[row, 1 : 1 + r_hat] = idx.float()
                 if pr is None:
-                    st[row, 17 : 17 + r_hat] = 1.0
+                    st[row, 1 + R_CAP : 1 + R_CAP + r_hat] = 1.0
                 else:
-                    st[row, 17 : 17 + r_hat] = pr.float()
+                    st[row, 1 + R_CAP : 1 + R_CAP + r_hat] = pr.float()
             self.sel_table_dev.copy_(self.sel_table_host, non_blocking=True)
 
         # host layers: build staged factors
@@ -385,7 +407,7 @@ class BatchedSVDEncoder:
             )
             e.batched_sel(
                 flat_grad, wire, self.stage_dev, self.desc, self.sel_work,
-                self.sel_work.shape[0],
+                self.sel_work.shape[0], self.sel_elems,
             )
         sd = self.stage_dev
         for i, r_hat in host_plans:
