@@ -28,8 +28,6 @@
 #include <cstdint>
 
 #define GD_N 8
-#define JMAX 64
-#define JSTRIDE 65
 #define SWEEPS 10
 #define JTHREADS 256
 #define JBIG_THREADS 512
@@ -54,10 +52,12 @@ __device__ __forceinline__ void rot_params(float app, float aqq, float apq,
 // ---------------------------------------------------------------------------
 // small variant: LDS-resident
 // ---------------------------------------------------------------------------
+template <int JMAX>
 __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
     float* __restrict__ grams, float* __restrict__ evals,
     const int64_t* __restrict__ desc, const int64_t* __restrict__ eval_offs,
     const int32_t* __restrict__ rows_list, int n_mats) {
+  constexpr int JSTRIDE = JMAX + 1;
   __shared__ float G[JMAX * JSTRIDE];
   __shared__ float V[JMAX * JSTRIDE];
   __shared__ float cs[JMAX / 2], sn[JMAX / 2];
@@ -362,10 +362,16 @@ extern "C" {
 
 void atomo_jacobi_eigh_launch(float* grams, float* evals, const int64_t* desc,
                               const int64_t* eval_offs,
-                              const int32_t* rows_list, int n_mats,
+                              const int32_t* rows_list, int n_mats, int jmax,
                               hipStream_t stream) {
-  hipLaunchKernelGGL(jacobi_eigh_kernel, dim3(n_mats), dim3(JTHREADS), 0,
-                     stream, grams, evals, desc, eval_offs, rows_list, n_mats);
+  if (jmax <= 64)
+    hipLaunchKernelGGL(jacobi_eigh_kernel<64>, dim3(n_mats), dim3(JTHREADS),
+                       0, stream, grams, evals, desc, eval_offs, rows_list,
+                       n_mats);
+  else
+    hipLaunchKernelGGL(jacobi_eigh_kernel<128>, dim3(n_mats), dim3(JTHREADS),
+                       0, stream, grams, evals, desc, eval_offs, rows_list,
+                       n_mats);
 }
 
 void atomo_jacobi_eigh_big_launch(float* grams, float* vbuf, float* evals,

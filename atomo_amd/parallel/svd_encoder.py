@@ -38,8 +38,13 @@ GRAM_CHUNK = 256
 SEL_CHUNK = 1024
 R_CAP = 32
 SEL_ROW = 1 + 2 * R_CAP  # [r_hat | idx*32 | probs*32]
-SMALL_SM = 64
+SMALL_SM = 64      # LDS-Jacobi (64-variant) + batched_gram kernel
+J128_SM = 128      # LDS-Jacobi (128-variant), Gram via rocBLAS
+# sm > J128_SM goes to the host LAPACK path unless ATOMO_BIG_JACOBI=1
+# (the global-memory Jacobi variant is latency-bound and loses to pooled
+# host eigh for the few big folds a model has).
 BIG_SM = 512
+_USE_BIG_JACOBI = os.environ.get("ATOMO_BIG_JACOBI", "0") not in ("", "0")
 
 
 class BatchedSVDEncoder:
@@ -98,15 +103,16 @@ class BatchedSVDEncoder:
             self.use_kernels = ops.have_ext()
         if self.use_kernels:
             desc_rows, gram_work, sel_work, eval_offs = [], [], [], []
-            small_rows, big_rows, big_v_offs = [], [], []
+            rows_j64, rows_j128, big_rows, big_v_offs = [], [], [], []
             ev_off, v_off, sel_elems = 0, 0, 1
+            sm_cap = BIG_SM if _USE_BIG_JACOBI else J128_SM
             for i, s in enumerate(specs):
                 m, n = s.meta["m"], s.meta["n"]
                 sm, tall = self.small[i], self.tall[i]
                 device_ok = (
                     s.meta["padded"] == s.numel
                     and s.meta["r_max"] <= R_CAP
-                    and sm <= BIG_SM
+                    and sm <= sm_cap
                     and (sm <= SMALL_SM or sm % 2 == 0)
                 )
                 if not device_ok:
@@ -130,9 +136,12 @@ class BatchedSVDEncoder:
                 self.kernel_rows.append(i)
                 sel_elems = max(sel_elems, sm * min(R_CAP, s.meta["r_max"]))
                 if sm <= SMALL_SM:
-                    small_rows.append(row)
+                    rows_j64.append(row)
                     for c in range((tall + GRAM_CHUNK - 1) // GRAM_CHUNK):
                         gram_work.append([row, c])
+                elif sm <= J128_SM:
+                    rows_j128.append(row)
+                    self.big_gram_layers.append(i)
                 else:
                     big_rows.append(row)
                     big_v_offs.append(v_off)
@@ -145,7 +154,8 @@ class BatchedSVDEncoder:
                 self.desc = torch.tensor(desc_rows, dtype=torch.int64, device=dev)
                 self.gram_work = torch.tensor(gram_work, dtype=torch.int32, device=dev)
                 self.sel_work = torch.tensor(sel_work, dtype=torch.int32, device=dev)
-                self.small_rows = torch.tensor(small_rows, dtype=torch.int32, device=dev)
+                self.rows_j64 = torch.tensor(rows_j64, dtype=torch.int32, device=dev)
+                self.rows_j128 = torch.tensor(rows_j128, dtype=torch.int32, device=dev)
                 self.big_rows = torch.tensor(big_rows, dtype=torch.int32, device=dev)
                 self.big_v_offs = torch.tensor(big_v_offs, dtype=torch.int64, device=dev)
                 self.vbuf = torch.zeros(max(1, v_off), dtype=torch.float32, device=dev)
@@ -281,13 +291,18 @@ class BatchedSVDEncoder:
                     gram_event.record()
                 e.jacobi_eigh(
                     self.grams, self.evals_dev, self.desc, self.eval_offs_dev,
-                    self.small_rows, self.small_rows.shape[0],
+                    self.rows_j64, self.rows_j64.shape[0], 64,
                 )
-                e.jacobi_eigh_big(
-                    self.grams, self.vbuf, self.evals_dev, self.desc,
-                    self.eval_offs_dev, self.big_rows, self.big_v_offs,
-                    self.big_rows.shape[0],
+                e.jacobi_eigh(
+                    self.grams, self.evals_dev, self.desc, self.eval_offs_dev,
+                    self.rows_j128, self.rows_j128.shape[0], 128,
                 )
+                if self.big_rows.shape[0]:
+                    e.jacobi_eigh_big(
+                        self.grams, self.vbuf, self.evals_dev, self.desc,
+                        self.eval_offs_dev, self.big_rows, self.big_v_offs,
+                        self.big_rows.shape[0],
+                    )
                 self.evals_host.copy_(self.evals_dev, non_blocking=True)
             elif host_layers:
                 grams_host = self.grams.to("cpu")  # synchronous copy
