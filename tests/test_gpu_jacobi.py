@@ -24,8 +24,8 @@ def _run_jacobi(sm, B=5, seed=None):
         desc[b] = torch.tensor([0, sm + 7, sm, 1, b * sm * sm, 0, 0, 8])
         eval_offs[b] = b * sm
     evals = torch.zeros(B * sm, device=dev)
-    if sm <= 64:
-        ext().jacobi_eigh(grams, evals, desc, eval_offs, rows, B)
+    if sm <= 128:
+        ext().jacobi_eigh(grams, evals, desc, eval_offs, rows, B, 64 if sm <= 64 else 128)
     else:
         v_offs = torch.tensor(
             [b * sm * sm for b in range(B)], dtype=torch.int64, device=dev
