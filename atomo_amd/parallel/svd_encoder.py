@@ -39,14 +39,12 @@ SEL_CHUNK = 1024
 R_CAP = 32
 SEL_ROW = 1 + 2 * R_CAP  # [r_hat | idx*32 | probs*32]
 SMALL_SM = 64      # LDS-Jacobi (64-variant) + batched_gram kernel
-# Bigger folds go to the pooled host LAPACK path, which overlaps the device
-# Jacobi: a 128/256 fp32 eigh is 0.9/3.2 ms on a hidden host thread, while
-# a device Jacobi of the same size sits on the evals-sync critical path.
-# ATOMO_JACOBI_CAP=128 / ATOMO_BIG_JACOBI=1 route them on device instead
-# (measured slower; kept for experiments).
+# 64 < sm <= J128_SM uses the LDS-128 Jacobi variant when enabled via
+# ATOMO_JACOBI_CAP=128 (measured slower than hipSOLVER on the sync path);
+# sm > that goes to batched hipSOLVER syevd on device (ResNet-50/152 have
+# 1x1-conv folds up to 2048x1024 — host LAPACK costs 30-60 ms each there).
 J128_SM = int(os.environ.get("ATOMO_JACOBI_CAP", "64"))
-BIG_SM = 512
-_USE_BIG_JACOBI = os.environ.get("ATOMO_BIG_JACOBI", "0") not in ("", "0")
+SOLVER_SM = 4096
 
 
 class BatchedSVDEncoder:
@@ -105,17 +103,17 @@ class BatchedSVDEncoder:
             self.use_kernels = ops.have_ext()
         if self.use_kernels:
             desc_rows, gram_work, sel_work, eval_offs = [], [], [], []
-            rows_j64, rows_j128, big_rows, big_v_offs = [], [], [], []
-            ev_off, v_off, sel_elems = 0, 0, 1
-            sm_cap = BIG_SM if _USE_BIG_JACOBI else J128_SM
+            rows_j64, rows_j128 = [], []
+            self.solver_layers = []  # big folds solved by hipSOLVER
+            ev_off, sel_elems = 0, 1
             for i, s in enumerate(specs):
                 m, n = s.meta["m"], s.meta["n"]
                 sm, tall = self.small[i], self.tall[i]
                 device_ok = (
                     s.meta["padded"] == s.numel
                     and s.meta["r_max"] <= R_CAP
-                    and sm <= sm_cap
-                    and (sm <= SMALL_SM or sm % 2 == 0)
+                    and sm <= SOLVER_SM
+                    and (sm <= J128_SM or sm % 2 == 0)
                 )
                 if not device_ok:
                     continue
@@ -145,9 +143,7 @@ class BatchedSVDEncoder:
                     rows_j128.append(row)
                     self.big_gram_layers.append(i)
                 else:
-                    big_rows.append(row)
-                    big_v_offs.append(v_off)
-                    v_off += sm * sm
+                    self.solver_layers.append(i)
                     self.big_gram_layers.append(i)
                 for c in range((tall + SEL_CHUNK - 1) // SEL_CHUNK):
                     sel_work.append([row, c])
@@ -158,9 +154,7 @@ class BatchedSVDEncoder:
                 self.sel_work = torch.tensor(sel_work, dtype=torch.int32, device=dev)
                 self.rows_j64 = torch.tensor(rows_j64, dtype=torch.int32, device=dev)
                 self.rows_j128 = torch.tensor(rows_j128, dtype=torch.int32, device=dev)
-                self.big_rows = torch.tensor(big_rows, dtype=torch.int32, device=dev)
-                self.big_v_offs = torch.tensor(big_v_offs, dtype=torch.int64, device=dev)
-                self.vbuf = torch.zeros(max(1, v_off), dtype=torch.float32, device=dev)
+                self.layer_row = {i: r for r, i in enumerate(self.kernel_rows)}
                 self.eval_offs_dev = torch.tensor(eval_offs, dtype=torch.int64, device=dev)
                 self.eval_offs = eval_offs
                 self.evals_dev = torch.zeros(max(1, ev_off), dtype=torch.float32, device=dev)
@@ -295,16 +289,39 @@ class BatchedSVDEncoder:
                     self.grams, self.evals_dev, self.desc, self.eval_offs_dev,
                     self.rows_j64, self.rows_j64.shape[0], 64,
                 )
-                e.jacobi_eigh(
-                    self.grams, self.evals_dev, self.desc, self.eval_offs_dev,
-                    self.rows_j128, self.rows_j128.shape[0], 128,
-                )
-                if self.big_rows.shape[0]:
-                    e.jacobi_eigh_big(
-                        self.grams, self.vbuf, self.evals_dev, self.desc,
-                        self.eval_offs_dev, self.big_rows, self.big_v_offs,
-                        self.big_rows.shape[0],
+                if self.rows_j128.shape[0]:
+                    e.jacobi_eigh(
+                        self.grams, self.evals_dev, self.desc,
+                        self.eval_offs_dev, self.rows_j128,
+                        self.rows_j128.shape[0], 128,
                     )
+                # big folds: batched hipSOLVER syevd, results written back
+                # into the gram slots / evals buffer on device
+                if self.solver_layers:
+                    by_sm = defaultdict(list)
+                    for i in self.solver_layers:
+                        by_sm[self.small[i]].append(i)
+                    for sm, idxs in by_sm.items():
+                        gs = torch.stack(
+                            [
+                                self.grams[
+                                    self.gram_offsets[i] : self.gram_offsets[i]
+                                    + sm * sm
+                                ].view(sm, sm)
+                                for i in idxs
+                            ]
+                        )
+                        gs = 0.5 * (gs + gs.transpose(1, 2))
+                        evals, evecs = torch.linalg.eigh(gs)
+                        evals = evals.flip(1).clamp(min=0.0)
+                        evecs = evecs.flip(2)
+                        for j, i in enumerate(idxs):
+                            o = self.eval_offs[self.layer_row[i]]
+                            self.evals_dev[o : o + sm].copy_(evals[j])
+                            self.grams[
+                                self.gram_offsets[i] : self.gram_offsets[i]
+                                + sm * sm
+                            ].copy_(evecs[j].reshape(-1))
                 self.evals_host.copy_(self.evals_dev, non_blocking=True)
             elif host_layers:
                 grams_host = self.grams.to("cpu")  # synchronous copy
