@@ -46,6 +46,18 @@ SMALL_SM = 64      # LDS-Jacobi (64-variant) + batched_gram kernel
 J128_SM = int(os.environ.get("ATOMO_JACOBI_CAP", "64"))
 SOLVER_SM = 4096
 
+# measured per-matrix costs (ms) on MI355X + EPYC host, fp32
+_HOST_EIGH_MS = {128: 0.9, 256: 3.3, 512: 10.7, 1024: 45.0, 2048: 160.0}
+_SOLVER_EIGH_MS = {128: 2.1, 256: 5.6, 512: 11.5, 1024: 21.0, 2048: 44.0}
+
+
+def _interp_cost(table, sm):
+    ks = sorted(table)
+    for k in ks:
+        if sm <= k:
+            return table[k]
+    return table[ks[-1]] * (sm / ks[-1]) ** 3
+
 
 class BatchedSVDEncoder:
     def __init__(
@@ -106,6 +118,23 @@ class BatchedSVDEncoder:
             rows_j64, rows_j128 = [], []
             self.solver_layers = []  # big folds solved by hipSOLVER
             ev_off, sel_elems = 0, 1
+            # group-level routing for big folds: batched hipSOLVER syevd is
+            # nearly count-free per call, pooled host LAPACK wins for one or
+            # two small-ish matrices (it overlaps the device Jacobi)
+            counts = defaultdict(int)
+            for i, s in enumerate(specs):
+                if self.small[i] > J128_SM:
+                    counts[self.small[i]] += 1
+            solver_dims = set()
+            for sm, cnt in counts.items():
+                host_ms = _interp_cost(_HOST_EIGH_MS, sm) * ((cnt + 7) // 8)
+                if cnt >= 3:
+                    host_ms = max(
+                        host_ms, _interp_cost(_HOST_EIGH_MS, sm) * cnt / 8.0
+                    )
+                solver_ms = _interp_cost(_SOLVER_EIGH_MS, sm)
+                if solver_ms < host_ms + 2.0 or sm >= 768:
+                    solver_dims.add(sm)
             for i, s in enumerate(specs):
                 m, n = s.meta["m"], s.meta["n"]
                 sm, tall = self.small[i], self.tall[i]
@@ -113,7 +142,7 @@ class BatchedSVDEncoder:
                     s.meta["padded"] == s.numel
                     and s.meta["r_max"] <= R_CAP
                     and sm <= SOLVER_SM
-                    and (sm <= J128_SM or sm % 2 == 0)
+                    and (sm <= J128_SM or sm in solver_dims)
                 )
                 if not device_ok:
                     continue
