@@ -133,7 +133,9 @@ class BatchedSVDEncoder:
                         host_ms, _interp_cost(_HOST_EIGH_MS, sm) * cnt / 8.0
                     )
                 solver_ms = _interp_cost(_SOLVER_EIGH_MS, sm)
-                if solver_ms < host_ms + 2.0 or sm >= 768:
+                # pooled host solves overlap ~2 ms of device work for free
+                host_eff = max(0.0, host_ms - 2.0)
+                if solver_ms < host_eff or sm >= 768:
                     solver_dims.add(sm)
             for i, s in enumerate(specs):
                 m, n = s.meta["m"], s.meta["n"]
