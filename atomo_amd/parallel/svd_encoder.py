@@ -127,11 +127,8 @@ class BatchedSVDEncoder:
                     counts[self.small[i]] += 1
             solver_dims = set()
             for sm, cnt in counts.items():
-                host_ms = _interp_cost(_HOST_EIGH_MS, sm) * ((cnt + 7) // 8)
-                if cnt >= 3:
-                    host_ms = max(
-                        host_ms, _interp_cost(_HOST_EIGH_MS, sm) * cnt / 8.0
-                    )
+                # measured: concurrent MKL eighs barely parallelize (~2x)
+                host_ms = _interp_cost(_HOST_EIGH_MS, sm) * max(1.0, cnt / 2.0)
                 solver_ms = _interp_cost(_SOLVER_EIGH_MS, sm)
                 # pooled host solves overlap ~2 ms of device work for free
                 host_eff = max(0.0, host_ms - 2.0)
