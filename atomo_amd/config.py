@@ -67,6 +67,9 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
                    help="encode layers on a side stream as backward produces them")
     p.add_argument("--graph", action="store_true", default=False,
                    help="capture forward/backward in a hipGraph and replay")
+    p.add_argument("--step-timeout", type=float, default=0.0,
+                   help="abort the rank if a global step stalls this many "
+                        "seconds (straggler/hang watchdog; 0 = off)")
     return parser
 
 
@@ -114,6 +117,8 @@ class RunConfig:
             checkpoint_freq=a.checkpoint_freq,
             train_dir=a.train_dir,
             use_graph=a.graph,
+            overlap=a.overlap,
+            step_timeout=a.step_timeout,
         )
 
 

@@ -259,6 +259,7 @@ class BatchedSVDEncoder:
         grads: List[torch.Tensor],
         wire: torch.Tensor,
         flat_grad: Optional[torch.Tensor] = None,
+        grams_done: bool = False,
     ) -> int:
         use_kernels = self.use_kernels and flat_grad is not None
         kernel_set = self.kernel_set if use_kernels else set()
@@ -281,13 +282,14 @@ class BatchedSVDEncoder:
             from ..ops import ext
 
             e = ext()
-            self.grams.zero_()
-            if self.gram_work.shape[0]:
-                e.batched_gram(
-                    flat_grad, self.grams, self.desc, self.gram_work,
-                    self.gram_work.shape[0],
-                )
-        for i in mm_layers:
+            if not grams_done:
+                self.grams.zero_()
+                if self.gram_work.shape[0]:
+                    e.batched_gram(
+                        flat_grad, self.grams, self.desc, self.gram_work,
+                        self.gram_work.shape[0],
+                    )
+        for i in (() if grams_done else mm_layers):
             a = a2ds[i]
             sm = self.small[i]
             gv = self.grams[self.gram_offsets[i] : self.gram_offsets[i] + sm * sm].view(

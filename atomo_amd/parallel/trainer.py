@@ -59,6 +59,8 @@ class PSTrainer:
         checkpoint_freq: int = 0,
         train_dir: str = "output/models/",
         use_graph: bool = False,
+        overlap: bool = False,
+        step_timeout: float = 0.0,
     ):
         self.comm = comm
         self.device = device or comm.device
@@ -127,6 +129,13 @@ class PSTrainer:
         self.train_dir = train_dir
         self.timers = PhaseTimers()
         self.last_loss = float("nan")
+        self.watchdog = None
+        if step_timeout and step_timeout > 0:
+            from ..utils.watchdog import StepWatchdog
+
+            self.watchdog = StepWatchdog(
+                step_timeout, action="abort", rank=comm.rank
+            ).start()
         # hipGraph capture of forward+backward: the per-step launch storm of
         # a convnet (hundreds of kernels) replays as one graph.  Input/label
         # buffers are static; flat_grad.zero_() is captured inside, so
@@ -136,6 +145,20 @@ class PSTrainer:
         self._static_x = None
         self._static_y = None
         self._static_loss = None
+        # backward-hook overlap (reference *Split capability): per-layer
+        # encode work on a side stream while backward continues.
+        self.overlap = (
+            bool(overlap)
+            and self.device.type == "cuda"
+            and self.is_worker
+            and not self.use_graph
+        )
+        self._side_stream = None
+        if self.overlap:
+            self._side_stream = torch.cuda.Stream()
+            self.overlap = self.wc.setup_overlap(
+                self._side_stream, self.params, self.wire
+            )
 
         # make every rank start from rank-0's init
         self.comm.broadcast(self.flat, src=0)
@@ -157,11 +180,17 @@ class PSTrainer:
                     loss = self.loss_fn(out, y)
                     loss.backward()
                     self.last_loss = float(loss.detach())
+                if self.overlap:
+                    torch.cuda.current_stream().wait_stream(self._side_stream)
             with t.phase("encode"):
                 if self.wc.reducible:
                     used = self.wc.total_words  # wire aliases flat_grad
                 else:
-                    used = self.wc.encode_all(self.wire, flat_grad=self.flat_grad)
+                    used = self.wc.encode_all(
+                        self.wire,
+                        flat_grad=self.flat_grad,
+                        overlap_done=self.overlap,
+                    )
                 t.add_scalar("msg_bytes", 4.0 * used)
         elif self.wc.reducible:
             self.flat_grad.zero_()  # dedicated PS contributes zeros to the sum
@@ -190,6 +219,8 @@ class PSTrainer:
                 self._apply(grad_flat)
 
         self.step_num += 1
+        if self.watchdog is not None:
+            self.watchdog.step()
         if self.step_num % self.shrink_freq == 0:
             self.lr *= self.lr_shrinkage
         if (

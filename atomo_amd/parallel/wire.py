@@ -61,15 +61,94 @@ class WireCodec:
     def reducible(self) -> bool:
         return getattr(self.codec, "reducible", False)
 
+    # -- overlap (backward-hook driven per-layer encode) -----------------
+    # The reference prototypes communication/computation overlap with its
+    # *Split models (per-layer backward interleaved with MPI Isend,
+    # model_ops/resnet_split.py:539-575).  Here the same capability is a
+    # post-accumulate-grad hook per parameter that launches that layer's
+    # encode work on a side HIP stream while backward continues on the
+    # deeper layers; encode_all(overlap_done=True) then skips the work.
+    def setup_overlap(self, side_stream, params, wire) -> bool:
+        import torch as _t
+
+        if self.device.type != "cuda":
+            return False
+        if isinstance(self.codec, SVDCodec) and self._batched_encoder is not None:
+            enc = self._batched_encoder
+
+            def make_hook(i):
+                spec = self.specs[i]
+                sm = enc.small[i]
+                gv = enc.grams[
+                    enc.gram_offsets[i] : enc.gram_offsets[i] + sm * sm
+                ].view(sm, sm)
+                tall_is_m = enc.m_is_tall[i]
+
+                def hook(p):
+                    evt = _t.cuda.Event()
+                    evt.record()
+                    with _t.cuda.stream(side_stream):
+                        side_stream.wait_event(evt)
+                        a = enc._a2d(p.grad, spec)
+                        if tall_is_m:
+                            _t.mm(a.t(), a, out=gv)
+                        else:
+                            _t.mm(a, a.t(), out=gv)
+
+                return hook
+
+            for i, p in enumerate(params):
+                p.register_post_accumulate_grad_hook(make_hook(i))
+            return True
+        if isinstance(self.codec, QSGDCodec):
+            from ..ops import qsgd_ops
+
+            codec = self.codec
+
+            def make_hook(i):
+                spec = self.specs[i]
+                region = wire[spec.wire_offset : spec.wire_offset + spec.wire_words]
+
+                def hook(p):
+                    evt = _t.cuda.Event()
+                    evt.record()
+                    with _t.cuda.stream(side_stream):
+                        side_stream.wait_event(evt)
+                        qsgd_ops.pack_into(
+                            p.grad.reshape(-1),
+                            region,
+                            codec.bucket_size,
+                            codec.qlevel,
+                            codec.scheme,
+                        )
+
+                return hook
+
+            for i, p in enumerate(params):
+                p.register_post_accumulate_grad_hook(make_hook(i))
+            return True
+        return False
+
     # -- worker side -----------------------------------------------------
-    def encode_all(self, wire: torch.Tensor, flat_grad: torch.Tensor = None) -> int:
+    def encode_all(
+        self,
+        wire: torch.Tensor,
+        flat_grad: torch.Tensor = None,
+        overlap_done: bool = False,
+    ) -> int:
         """Encode every parameter's .grad into ``wire``; returns fp32 words
-        actually used (the Msg bytes counter)."""
+        actually used (the Msg bytes counter).  overlap_done=True means the
+        backward hooks already did the per-layer front half (QSGD: the whole
+        pack; SVD: the Gram matrices)."""
+        if overlap_done and isinstance(self.codec, QSGDCodec):
+            return sum(s.wire_words for s in self.specs)
         grads = [
             p.grad if p.grad is not None else torch.zeros_like(p) for p in self.params
         ]
         if self._batched_encoder is not None:
-            return self._batched_encoder.encode_all(grads, wire, flat_grad=flat_grad)
+            return self._batched_encoder.encode_all(
+                grads, wire, flat_grad=flat_grad, grams_done=overlap_done
+            )
         used = 0
         for grad, spec in zip(grads, self.specs):
             region = wire[spec.wire_offset : spec.wire_offset + spec.wire_words]

@@ -40,6 +40,8 @@ def main():
                    help="capture forward/backward in a hipGraph and replay")
     p.add_argument("--channels-last", action="store_true", default=False,
                    help="NHWC memory format for convs")
+    p.add_argument("--overlap", action="store_true", default=False,
+                   help="backward-hook per-layer encode on a side stream")
     a = p.parse_args()
 
     from atomo_amd.codings import make_codec
@@ -72,6 +74,7 @@ def main():
         seed=42,
         device=device,
         use_graph=a.graph,
+        overlap=a.overlap,
     )
     if a.channels_last:
         trainer.model.to(memory_format=torch.channels_last)

@@ -118,6 +118,32 @@ def test_gpu_raw_equals_plain_sgd(dev):
     assert diff < 2e-3 * max(1.0, scale), diff
 
 
+@pytest.mark.parametrize("code", ["svd", "qsgd"])
+def test_gpu_overlap_trains(code, dev):
+    """Backward-hook overlap path: per-layer encode on a side stream gives
+    the same training behavior (LeNet memorizes the pool)."""
+    from atomo_amd.codings import make_codec
+    from atomo_amd.data import make_loaders
+    from atomo_amd.parallel import Comm, PSTrainer
+
+    comm = Comm(device=dev)
+    trainer = PSTrainer(
+        model_name="LeNet",
+        codec=make_codec(code, rank=3, quantization_level=4, bucket_size=512),
+        comm=comm, lr=0.05, momentum=0.9, num_classes=10, in_channels=1,
+        seed=11, device=dev, overlap=True,
+    )
+    assert trainer.overlap, "overlap hooks failed to install"
+    train, _ = make_loaders("mnist", 32, 32, dev, seed=5)
+    it = iter_cycle(train)
+    losses = []
+    for _ in range(30):
+        x, y = next(it)
+        losses.append(trainer.train_step(x, y))
+    assert all(not math.isnan(l) for l in losses)
+    assert sum(losses[-5:]) / 5 < sum(losses[:5]) / 5, losses
+
+
 def test_gpu_svd_wire_unbiased(dev):
     """Whole wire round trip (encode_into + decode_acc kernel) is unbiased."""
     from atomo_amd.codings import SVDCodec
