@@ -43,6 +43,9 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
     p.add_argument("--code", type=str, default="svd", choices=["sgd", "svd", "qsgd"])
     p.add_argument("--bucket-size", type=int, default=512)
     p.add_argument("--dataset", type=str, default="cifar10")
+    p.add_argument("--data-root", type=str, default=None,
+                   help="directory with the real dataset files (standard "
+                        "binary formats); synthetic data when absent")
     p.add_argument("--comm-type", type=str, default="Bcast")
     p.add_argument("--num-aggregate", type=int, default=0,
                    help="gradients to collect per step (0 = all workers; the "

@@ -81,7 +81,18 @@ def make_loaders(
     test_batch_size: int,
     device: torch.device,
     seed: int = 0,
+    root: str = None,
 ):
+    """Synthetic loaders by default; real on-disk data when ``root`` holds
+    the dataset's standard files (atomo_amd.data.disk)."""
+    from .disk import DiskImageData, has_disk_data
+
+    if has_disk_data(dataset, root):
+        train = DiskImageData(dataset, root, batch_size, device, train=True,
+                              seed=seed)
+        test = DiskImageData(dataset, root, test_batch_size, device,
+                             train=False, seed=seed + 1)
+        return train, test
     train = SyntheticImageData(dataset, batch_size, device, seed=seed)
     test = SyntheticImageData(dataset, test_batch_size, device, seed=seed + 1,
                               pool_batches=4, batches_per_epoch=4)

@@ -32,7 +32,8 @@ def main(argv=None):
     trainer = PSTrainer(codec=codec, comm=comm, device=device, **cfg.trainer_kwargs())
 
     train_loader, test_loader = make_loaders(
-        a.dataset, a.batch_size, a.test_batch_size, device, seed=a.seed + comm.rank
+        a.dataset, a.batch_size, a.test_batch_size, device,
+        seed=a.seed + comm.rank, root=a.data_root
     )
 
     step, done = 0, False

@@ -22,7 +22,8 @@ def main(argv=None):
     codec = cfg.build_codec()
     trainer = PSTrainer(codec=codec, comm=comm, device=device, **cfg.trainer_kwargs())
     train_loader, test_loader = make_loaders(
-        a.dataset, a.batch_size, a.test_batch_size, device, seed=a.seed
+        a.dataset, a.batch_size, a.test_batch_size, device, seed=a.seed,
+        root=a.data_root
     )
     step = 0
     t0 = time.perf_counter()
