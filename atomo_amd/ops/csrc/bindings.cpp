@@ -31,6 +31,9 @@ void atomo_jacobi_eigh_big_launch(float*, float*, float*, const int64_t*,
 void atomo_build_stage_launch(const float*, const float*, const float*,
                               float*, const int64_t*, const int64_t*, int,
                               hipStream_t);
+void atomo_sample_stage_launch(const float*, const float*, float*,
+                               const int64_t*, const int64_t*, int, int, int,
+                               uint64_t, unsigned long long*, hipStream_t);
 }
 
 namespace {
@@ -197,9 +200,31 @@ void build_stage(torch::Tensor evecs, torch::Tensor evals,
       cur_stream());
 }
 
+void sample_stage(torch::Tensor evecs, torch::Tensor evals,
+                  torch::Tensor stage, torch::Tensor desc,
+                  torch::Tensor eval_offs, int64_t n_layers, int64_t rank,
+                  bool truncate, int64_t seed, torch::Tensor used_words) {
+  check_f32_cuda(evecs, "evecs");
+  check_f32_cuda(evals, "evals");
+  check_f32_cuda(stage, "stage");
+  TORCH_CHECK(used_words.is_cuda() &&
+                  used_words.scalar_type() == torch::kInt64,
+              "used_words must be cuda int64");
+  if (n_layers == 0) return;
+  atomo_sample_stage_launch(
+      evecs.data_ptr<float>(), evals.data_ptr<float>(),
+      stage.data_ptr<float>(), desc.data_ptr<int64_t>(),
+      eval_offs.data_ptr<int64_t>(), (int)n_layers, (int)rank,
+      truncate ? 1 : 0, (uint64_t)seed,
+      reinterpret_cast<unsigned long long*>(used_words.data_ptr<int64_t>()),
+      cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("sample_stage", &sample_stage,
+        "fused on-device Bernoulli atom sampler + stage builder");
   m.def("jacobi_eigh", &jacobi_eigh,
         "batched parallel-Jacobi symmetric eigensolver (sm <= 64, LDS)");
   m.def("jacobi_eigh_big", &jacobi_eigh_big,

@@ -356,9 +356,141 @@ __global__ void __launch_bounds__(64) build_stage_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// fused on-device sampler + stage builder: removes the host round trip.
+// Per layer (one wave64 WG): p_i = min(1, rank*s_i/sum s) (rank==0:
+// s_i/s_0), one counter-hash Bernoulli draw per atom, redraw-until-nonempty
+// (reference _sample_svd, codings/svd.py:49-67), cap at r_max keeping the
+// highest-probability atoms; then build the staged wire factors exactly
+// like build_stage_kernel.  used_words accumulates the actual packet sizes
+// (the Msg-bytes counter) on device.
+// ---------------------------------------------------------------------------
+namespace {
+
+__device__ __forceinline__ float u01_hash2(uint64_t seed, uint64_t idx) {
+  uint64_t z = seed + 0x9E3779B97F4A7C15ull * (idx + 1);
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  z = z ^ (z >> 31);
+  return (float)(z >> 40) * (1.0f / 16777216.0f);
+}
+
+#define SAMPLE_SM_MAX 4096
+
+__global__ void __launch_bounds__(64) sample_stage_kernel(
+    const float* __restrict__ evecs, const float* __restrict__ evals,
+    float* __restrict__ stage, const int64_t* __restrict__ desc,
+    const int64_t* __restrict__ eval_offs, int n_layers, int rank,
+    int truncate, uint64_t seed,
+    unsigned long long* __restrict__ used_words) {
+  __shared__ float s_lds[SAMPLE_SM_MAX];
+  __shared__ float s_sel[R_CAP], inv_s[R_CAP];
+  __shared__ int idxs[R_CAP];
+  __shared__ int r_hat_s;
+  const int layer = blockIdx.x;
+  if (layer >= n_layers) return;
+  const int64_t* d = desc + (int64_t)layer * GD_N;
+  const int m = (int)d[1], n = (int)d[2];
+  const bool is_tall = d[3] != 0;
+  const int sm = is_tall ? n : m;
+  const int64_t so = d[6];
+  const int r_max = (int)d[7];
+  const float* V = evecs + d[4];
+  const float* ev = evals + eval_offs[layer];
+  const int tid = threadIdx.x;
+
+  // s = sqrt(eval), staged in LDS; wave-sum
+  float ssum = 0.f;
+  for (int i = tid; i < sm; i += 64) {
+    const float s = sqrtf(fmaxf(ev[i], 0.f));
+    s_lds[i] = s;
+    ssum += s;
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) ssum += __shfl_xor(ssum, off, 64);
+
+  if (tid == 0) {
+    int r_hat = 0;
+    if (truncate) {  // deterministic top-r (master-style, svd.py:109-113)
+      int r = rank > 0 ? rank : r_max;
+      if (r > r_max) r = r_max;
+      if (r > sm) r = sm;
+      if (r > R_CAP) r = R_CAP;
+      for (int i = 0; i < r; ++i) {
+        idxs[i] = i;
+        s_sel[i] = s_lds[i];
+        inv_s[i] = s_lds[i] > 1e-12f ? 1.f / s_lds[i] : 0.f;
+      }
+      r_hat = r;
+    } else if (s_lds[0] < 1e-6f) {  // degenerate: ship atom 0 with p = 1
+      idxs[0] = 0;
+      s_sel[0] = s_lds[0];
+      inv_s[0] = 0.f;
+      r_hat = 1;
+    } else {
+      const float inv_sum = 1.0f / ssum;
+      for (int attempt = 0; attempt < 64 && r_hat == 0; ++attempt) {
+        const uint64_t base =
+            seed + (uint64_t)layer * 0x100000ull + (uint64_t)attempt * 0x10000ull;
+        for (int i = 0; i < sm; ++i) {
+          float p = (rank == 0) ? s_lds[i] / s_lds[0]
+                                : (float)rank * s_lds[i] * inv_sum;
+          p = fminf(p, 1.0f);
+          if (u01_hash2(base, (uint64_t)i) < p) {
+            if (r_hat < r_max && r_hat < R_CAP) {
+              idxs[r_hat] = i;
+              const float s = s_lds[i];
+              s_sel[r_hat] = s / p;  // unbiased rescale
+              inv_s[r_hat] = s > 1e-12f ? 1.f / s : 0.f;
+              ++r_hat;
+            }
+            // overflow beyond the wire budget: drop (tail event; the
+            // highest-probability atoms come first since s is sorted)
+          }
+        }
+      }
+      if (r_hat == 0) {  // 64 failed redraws: ship the top atom, p ~ its prob
+        float p = (rank == 0) ? 1.0f : fminf((float)rank * s_lds[0] * inv_sum, 1.0f);
+        idxs[0] = 0;
+        s_sel[0] = s_lds[0] / p;
+        inv_s[0] = s_lds[0] > 1e-12f ? 1.f / s_lds[0] : 0.f;
+        r_hat = 1;
+      }
+    }
+    r_hat_s = r_hat;
+    stage[so] = (float)r_hat;
+    atomicAdd(used_words,
+              (unsigned long long)(1 + r_hat * (m + n + 1)));
+  }
+  __syncthreads();
+  const int r_hat = r_hat_s;
+  for (int k = tid; k < r_hat; k += 64) stage[so + 1 + k] = s_sel[k];
+  float* facT = stage + so + 1 + r_max;
+  float* sel = stage + so + 1 + (int64_t)r_max * (1 + sm);
+  for (int i = tid; i < r_hat * sm; i += 64) {
+    const int r = i / sm, k = i % sm;
+    const float v = V[(int64_t)k * sm + idxs[r]];
+    facT[(int64_t)r * sm + k] = v;
+    sel[(int64_t)k * r_hat + r] = v * inv_s[r];
+  }
+}
+
+}  // namespace (inner)
 }  // namespace
 
 extern "C" {
+
+void atomo_sample_stage_launch(const float* evecs, const float* evals,
+                               float* stage, const int64_t* desc,
+                               const int64_t* eval_offs, int n_layers,
+                               int rank, int truncate, uint64_t seed,
+                               unsigned long long* used_words,
+                               hipStream_t stream) {
+  hipLaunchKernelGGL(sample_stage_kernel, dim3(n_layers), dim3(64), 0, stream,
+                     evecs, evals, stage, desc, eval_offs, n_layers, rank,
+                     truncate, seed, used_words);
+}
+
 
 void atomo_jacobi_eigh_launch(float* grams, float* evals, const int64_t* desc,
                               const int64_t* eval_offs,

@@ -192,6 +192,8 @@ class BatchedSVDEncoder:
                 self.sel_table_dev = torch.zeros(nrows, SEL_ROW, dtype=torch.float32, device=dev)
                 self.grams_host = torch.zeros(gram_off, dtype=torch.float32, pin_memory=True)
                 self.sel_elems = sel_elems
+                self.used_words_dev = torch.zeros(1, dtype=torch.int64, device=dev)
+                self._seed = 0x9E3779B97F4A7C15
             else:
                 self.use_kernels = False
 
@@ -357,6 +359,25 @@ class BatchedSVDEncoder:
                 grams_host = self.grams.to("cpu")  # synchronous copy
         else:
             grams_host = self.grams
+
+        # ---- fully-async path: on-device sampler, no host sync ---------
+        # (every layer on device, RNG not pinned to a host generator)
+        if use_kernels and not host_layers and self.codec.generator is None:
+            self._seed = (self._seed * 6364136223846793005 + 1442695040888963407) % (
+                1 << 62
+            )
+            e.sample_stage(
+                self.grams, self.evals_dev, self.stage_dev, self.desc,
+                self.eval_offs_dev, len(self.kernel_rows),
+                self.codec.rank, not self.codec.random_sample, self._seed,
+                self.used_words_dev,
+            )
+            e.batched_sel(
+                flat_grad, wire, self.stage_dev, self.desc, self.sel_work,
+                self.sel_work.shape[0], self.sel_elems,
+            )
+            mark("async sample+sel")
+            return -1  # used words accumulate in used_words_dev
 
         def _solve_group(item):
             sm, idxs = item

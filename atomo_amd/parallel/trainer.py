@@ -61,6 +61,7 @@ class PSTrainer:
         use_graph: bool = False,
         overlap: bool = False,
         step_timeout: float = 0.0,
+        defer_loss: bool = False,
     ):
         self.comm = comm
         self.device = device or comm.device
@@ -129,6 +130,10 @@ class PSTrainer:
         self.train_dir = train_dir
         self.timers = PhaseTimers()
         self.last_loss = float("nan")
+        # defer_loss: keep the loss on device (no per-step host sync); read
+        # it via current_loss() when logging.
+        self.defer_loss = bool(defer_loss)
+        self._loss_tensor = None
         self.watchdog = None
         if step_timeout and step_timeout > 0:
             from ..utils.watchdog import StepWatchdog
@@ -179,7 +184,10 @@ class PSTrainer:
                     out = self.model(x)
                     loss = self.loss_fn(out, y)
                     loss.backward()
-                    self.last_loss = float(loss.detach())
+                    if self.defer_loss:
+                        self._loss_tensor = loss.detach()
+                    else:
+                        self.last_loss = float(loss.detach())
                 if self.overlap:
                     torch.cuda.current_stream().wait_stream(self._side_stream)
             with t.phase("encode"):
@@ -191,7 +199,8 @@ class PSTrainer:
                         flat_grad=self.flat_grad,
                         overlap_done=self.overlap,
                     )
-                t.add_scalar("msg_bytes", 4.0 * used)
+                if used >= 0:
+                    t.add_scalar("msg_bytes", 4.0 * used)
         elif self.wc.reducible:
             self.flat_grad.zero_()  # dedicated PS contributes zeros to the sum
 
@@ -229,6 +238,11 @@ class PSTrainer:
             and self.step_num % self.checkpoint_freq == 0
         ):
             self.save_checkpoint()
+        return self.last_loss
+
+    def current_loss(self) -> float:
+        if self._loss_tensor is not None:
+            self.last_loss = float(self._loss_tensor)
         return self.last_loss
 
     def _fwd_bwd_graphed(self, x: torch.Tensor, y: torch.Tensor) -> None:

@@ -61,6 +61,20 @@ class WireCodec:
     def reducible(self) -> bool:
         return getattr(self.codec, "reducible", False)
 
+    _device_counted = False
+
+    def device_msg_bytes(self) -> float:
+        """Msg bytes accumulated on device by the async sampler (one D2H)."""
+        if self._batched_encoder is None or not self._device_counted:
+            return 0.0
+        return 4.0 * float(self._batched_encoder.used_words_dev.item())
+
+    def reset_device_msg_bytes(self) -> None:
+        if self._batched_encoder is not None and hasattr(
+            self._batched_encoder, "used_words_dev"
+        ):
+            self._batched_encoder.used_words_dev.zero_()
+
     # -- overlap (backward-hook driven per-layer encode) -----------------
     # The reference prototypes communication/computation overlap with its
     # *Split models (per-layer backward interleaved with MPI Isend,
@@ -146,9 +160,12 @@ class WireCodec:
             p.grad if p.grad is not None else torch.zeros_like(p) for p in self.params
         ]
         if self._batched_encoder is not None:
-            return self._batched_encoder.encode_all(
+            used = self._batched_encoder.encode_all(
                 grads, wire, flat_grad=flat_grad, grams_done=overlap_done
             )
+            if used < 0:
+                self._device_counted = True
+            return used
         used = 0
         for grad, spec in zip(grads, self.specs):
             region = wire[spec.wire_offset : spec.wire_offset + spec.wire_words]

@@ -54,7 +54,8 @@ def test_batched_kernels_match_truncated_svd(dev):
     from atomo_amd.codings.svd import grad_to_2d
 
     used = enc.encode_all(grads, wire, flat_grad=flat)
-    assert used > 0
+    # -1 = fully-async device path (used words accumulate on device)
+    assert used == -1 or used > 0
     for g, spec in zip(grads, specs):
         region = wire[spec.wire_offset : spec.wire_offset + spec.wire_words].cpu()
         out = torch.zeros(spec.numel)
