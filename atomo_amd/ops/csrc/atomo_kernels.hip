@@ -172,6 +172,140 @@ __global__ void qsgd_unpack_acc_kernel(const float* __restrict__ norms,
 }
 
 // ---------------------------------------------------------------------------
+// Batched QSGD over ALL layers in one launch (descriptor-table driven, like
+// the SVD-encode kernels).  Descriptor row (int64, QD_N words):
+//   [0] grad_off (into flat_grad)  [1] numel  [2] wire_off  [3] n_buckets
+//   [4] words_per_bucket
+// pack work map: (layer, local bucket) per wave; unpack work map:
+// (layer, 256-word chunk) per workgroup.
+// ---------------------------------------------------------------------------
+#define QD_N 5
+
+__global__ void qsgd_pack_batched_kernel(
+    const float* __restrict__ flat, float* __restrict__ wire,
+    const int64_t* __restrict__ desc, const int32_t* __restrict__ work,
+    int n_tiles, int bucket_size, int qlevel, bool terngrad, uint64_t seed) {
+  extern __shared__ uint8_t lds_codes[];
+  const int waves_per_wg = blockDim.x / WAVE;
+  const int wave_id = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int bits = 1 + qlevel;
+  const int epw = 32 / bits;
+  const int s_levels = (1 << qlevel) - 1;
+  uint8_t* my_codes = lds_codes + (size_t)wave_id * bucket_size;
+
+  for (int tile = blockIdx.x * waves_per_wg + wave_id;
+       tile < n_tiles; tile += gridDim.x * waves_per_wg) {
+    const int layer = work[2 * tile];
+    const int bucket = work[2 * tile + 1];
+    const int64_t* d = desc + (int64_t)layer * QD_N;
+    const float* grad = flat + d[0];
+    const int64_t numel = d[1];
+    const int wpb = (int)d[4];
+    float* norms = wire + d[2];
+    uint32_t* packed = reinterpret_cast<uint32_t*>(norms + d[3]);
+    const int64_t base = (int64_t)bucket * bucket_size;
+    float ssq = 0.f, sum = 0.f;
+    for (int e = lane; e < bucket_size; e += WAVE) {
+      const int64_t g = base + e;
+      const float w = (g < numel) ? grad[g] : 0.f;
+      ssq += w * w;
+      sum += w;
+    }
+    float norm, limit = 0.f;
+    if (terngrad) {
+      const float n_inv = 1.0f / (float)bucket_size;
+      const float mean = wave_reduce_sum(sum) * n_inv;
+      const float var = fmaxf(wave_reduce_sum(ssq) * n_inv - mean * mean, 0.f);
+      limit = 2.5f * sqrtf(var);
+      float cmax = 0.f;
+      for (int e = lane; e < bucket_size; e += WAVE) {
+        const int64_t g = base + e;
+        float w = (g < numel) ? grad[g] : 0.f;
+        w = fminf(fmaxf(w, -limit), limit);
+        cmax = fmaxf(cmax, fabsf(w));
+      }
+      norm = wave_reduce_max(cmax);
+    } else {
+      norm = sqrtf(wave_reduce_sum(ssq));
+    }
+    if (lane == 0) norms[bucket] = norm;
+    const float inv_norm_s = (norm > 1e-30f) ? ((float)s_levels / norm) : 0.f;
+    for (int e = lane; e < bucket_size; e += WAVE) {
+      const int64_t g = base + e;
+      float w = (g < numel) ? grad[g] : 0.f;
+      if (terngrad) w = fminf(fmaxf(w, -limit), limit);
+      const float scaled = fabsf(w) * inv_norm_s;
+      int xi = (int)scaled;
+      const float frac = scaled - (float)xi;
+      if (u01_hash(seed + (uint64_t)layer * 0x1000000000000ull, (uint64_t)g) <
+          frac)
+        xi += 1;
+      if (xi > s_levels) xi = s_levels;
+      my_codes[e] = (uint8_t)((((w < 0.f) ? 1 : 0) << qlevel) | xi);
+    }
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "workgroup");
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "workgroup");
+    uint32_t* out = packed + (int64_t)bucket * wpb;
+    for (int wrd = lane; wrd < wpb; wrd += WAVE) {
+      uint32_t acc = 0;
+      const int e0 = wrd * epw;
+      for (int k = 0; k < epw; ++k) {
+        const int e = e0 + k;
+        const uint32_t c = (e < bucket_size) ? my_codes[e] : 0u;
+        acc |= c << (k * bits);
+      }
+      out[wrd] = acc;
+    }
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "workgroup");
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "workgroup");
+  }
+}
+
+#define QUNPACK_CHUNK 256
+
+__global__ void qsgd_unpack_batched_kernel(
+    const float* __restrict__ wire, float* __restrict__ agg,
+    const int64_t* __restrict__ desc, const int32_t* __restrict__ work,
+    int n_tiles, int bucket_size, int qlevel) {
+  const int bits = 1 + qlevel;
+  const int epw = 32 / bits;
+  const int s_levels = (1 << qlevel) - 1;
+  const uint32_t ximask = (1u << qlevel) - 1u;
+  for (int tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+    const int layer = work[2 * tile];
+    const int chunk = work[2 * tile + 1];
+    const int64_t* d = desc + (int64_t)layer * QD_N;
+    const int64_t numel = d[1];
+    const int nb = (int)d[3];
+    const int wpb = (int)d[4];
+    const float* norms = wire + d[2];
+    const uint32_t* packed = reinterpret_cast<const uint32_t*>(norms + nb);
+    float* out = agg + d[0];
+    const int64_t n_words = (int64_t)nb * wpb;
+    const int64_t w0 = (int64_t)chunk * QUNPACK_CHUNK;
+    const int64_t w1 = min(w0 + QUNPACK_CHUNK, n_words);
+    for (int64_t w = w0 + threadIdx.x; w < w1; w += blockDim.x) {
+      const int bucket = (int)(w / wpb);
+      const int wib = (int)(w % wpb);
+      const float scale = norms[bucket] / (float)s_levels;
+      const uint32_t word = packed[w];
+      const int64_t ebase =
+          (int64_t)bucket * bucket_size + (int64_t)wib * epw;
+      for (int k = 0; k < epw; ++k) {
+        const uint32_t c = (word >> (k * bits)) & ((1u << bits) - 1u);
+        const int64_t e = ebase + k;
+        if (e < numel && (int64_t)wib * epw + k < bucket_size) {
+          const float xi = (float)(c & ximask);
+          const float sgn = (c >> qlevel) ? -1.f : 1.f;
+          out[e] += sgn * xi * scale;
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // SVD decode + accumulate over W packets in one output sweep.
 // Packet layout (fp32): [r_hat | uT (r_max x m) | s (r_max) | vT (r_max x n)]
 // out2d(m, n) += sum_w sum_r u_w[i,r] * s_w[r] * vT_w[r,j]
@@ -264,6 +398,32 @@ void atomo_svd_decode_acc_launch(const float* regions, float* out, int W,
   const int grid = grid_for((int64_t)m * n, block);
   hipLaunchKernelGGL(svd_decode_acc_kernel, dim3(grid), dim3(block), 0, stream,
                      regions, out, W, stride, m, n, r_max);
+}
+
+void atomo_qsgd_pack_batched_launch(const float* flat, float* wire,
+                                    const int64_t* desc, const int32_t* work,
+                                    int n_tiles, int bucket_size, int qlevel,
+                                    bool terngrad, uint64_t seed,
+                                    hipStream_t stream) {
+  const int block = 256;
+  const int waves_per_wg = block / WAVE;
+  const size_t lds = (size_t)waves_per_wg * bucket_size;
+  const int grid = grid_for((int64_t)n_tiles * WAVE, block);
+  hipLaunchKernelGGL(qsgd_pack_batched_kernel, dim3(grid), dim3(block), lds,
+                     stream, flat, wire, desc, work, n_tiles, bucket_size,
+                     qlevel, terngrad, seed);
+}
+
+void atomo_qsgd_unpack_batched_launch(const float* wire, float* agg,
+                                      const int64_t* desc,
+                                      const int32_t* work, int n_tiles,
+                                      int bucket_size, int qlevel,
+                                      hipStream_t stream) {
+  const int grid = grid_for(n_tiles, 1) > 8192 ? 8192
+                   : (n_tiles < 1 ? 1 : n_tiles);
+  hipLaunchKernelGGL(qsgd_unpack_batched_kernel, dim3(grid), dim3(256), 0,
+                     stream, wire, agg, desc, work, n_tiles, bucket_size,
+                     qlevel);
 }
 
 void atomo_fused_sgd_launch(float* p, const float* g, float* buf, int64_t n,

@@ -18,6 +18,12 @@ void atomo_svd_decode_acc_launch(const float*, float*, int, int64_t, int, int,
                                  int, hipStream_t);
 void atomo_fused_sgd_launch(float*, const float*, float*, int64_t, float,
                             float, float, bool, float, float, hipStream_t);
+void atomo_qsgd_pack_batched_launch(const float*, float*, const int64_t*,
+                                    const int32_t*, int, int, int, bool,
+                                    uint64_t, hipStream_t);
+void atomo_qsgd_unpack_batched_launch(const float*, float*, const int64_t*,
+                                      const int32_t*, int, int, int,
+                                      hipStream_t);
 void atomo_batched_gram_launch(const float*, float*, const int64_t*,
                                const int32_t*, int, hipStream_t);
 void atomo_batched_sel_launch(const float*, float*, const float*,
@@ -115,6 +121,34 @@ void fused_sgd(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
                          (float)lr, (float)momentum, (float)weight_decay,
                          nesterov, (float)dampening, (float)grad_scale,
                          cur_stream());
+}
+
+void qsgd_pack_batched(torch::Tensor flat, torch::Tensor wire,
+                       torch::Tensor desc, torch::Tensor work,
+                       int64_t n_tiles, int64_t bucket_size, int64_t qlevel,
+                       bool terngrad, int64_t seed) {
+  check_f32_cuda(flat, "flat");
+  check_f32_cuda(wire, "wire");
+  TORCH_CHECK(bucket_size <= 8192, "bucket_size too large for LDS staging");
+  if (n_tiles == 0) return;
+  atomo_qsgd_pack_batched_launch(
+      flat.data_ptr<float>(), wire.data_ptr<float>(),
+      desc.data_ptr<int64_t>(), work.data_ptr<int32_t>(), (int)n_tiles,
+      (int)bucket_size, (int)qlevel, terngrad, (uint64_t)seed, cur_stream());
+}
+
+void qsgd_unpack_batched(torch::Tensor wire, torch::Tensor agg,
+                         torch::Tensor desc, torch::Tensor work,
+                         int64_t n_tiles, int64_t bucket_size,
+                         int64_t qlevel) {
+  TORCH_CHECK(wire.is_cuda() && wire.scalar_type() == torch::kFloat32,
+              "wire must be cuda fp32");
+  check_f32_cuda(agg, "agg");
+  if (n_tiles == 0) return;
+  atomo_qsgd_unpack_batched_launch(
+      wire.data_ptr<float>(), agg.data_ptr<float>(), desc.data_ptr<int64_t>(),
+      work.data_ptr<int32_t>(), (int)n_tiles, (int)bucket_size, (int)qlevel,
+      cur_stream());
 }
 
 void batched_gram(torch::Tensor flat, torch::Tensor grams, torch::Tensor desc,
@@ -236,6 +270,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("batched_sel", &batched_sel,
         "batched selection GEMM + packet scatter into the wire");
   m.def("qsgd_pack", &qsgd_pack, "QSGD bucket quantize+pack (gfx950)");
+  m.def("qsgd_pack_batched", &qsgd_pack_batched,
+        "one-launch QSGD pack over all layers");
+  m.def("qsgd_unpack_batched", &qsgd_unpack_batched,
+        "one-launch QSGD unpack+accumulate over all layers");
   m.def("qsgd_unpack_acc", &qsgd_unpack_acc, "QSGD unpack+accumulate");
   m.def("svd_decode_acc", &svd_decode_acc,
         "fused rank-k SVD decode+accumulate over W packets");

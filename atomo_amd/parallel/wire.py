@@ -48,6 +48,31 @@ class WireCodec:
             self._batched_encoder = BatchedSVDEncoder(
                 codec, self.specs, device, param_offsets=self.param_offsets
             )
+        # batched QSGD tables: one pack / unpack launch for the whole model
+        self._qsgd_tables = None
+        if device.type == "cuda" and isinstance(codec, QSGDCodec):
+            from .. import ops
+
+            if ops.have_ext():
+                desc, pack_work, unpack_work = [], [], []
+                for spec, p_off in zip(self.specs, self.param_offsets):
+                    meta = spec.meta
+                    nb, wpb = meta["n_buckets"], meta["words_per_bucket"]
+                    li = len(desc)
+                    desc.append([p_off, spec.numel, spec.wire_offset, nb, wpb])
+                    for b in range(nb):
+                        pack_work.append([li, b])
+                    n_words = nb * wpb
+                    for c in range((n_words + 255) // 256):
+                        unpack_work.append([li, c])
+                self._qsgd_tables = {
+                    "desc": torch.tensor(desc, dtype=torch.int64, device=device),
+                    "pack": torch.tensor(pack_work, dtype=torch.int32, device=device),
+                    "unpack": torch.tensor(
+                        unpack_work, dtype=torch.int32, device=device
+                    ),
+                }
+                self._qsgd_seed = 987654321
         # scratch for layers whose 2-D fold is zero-padded (odd 1-D sizes)
         self._pad_scratch = {}
         if isinstance(codec, SVDCodec) and codec.compress:
@@ -156,6 +181,19 @@ class WireCodec:
         pack; SVD: the Gram matrices)."""
         if overlap_done and isinstance(self.codec, QSGDCodec):
             return sum(s.wire_words for s in self.specs)
+        if self._qsgd_tables is not None and flat_grad is not None:
+            from ..ops import ext
+
+            t = self._qsgd_tables
+            self._qsgd_seed = (
+                self._qsgd_seed * 6364136223846793005 + 1442695040888963407
+            ) % (1 << 62)
+            ext().qsgd_pack_batched(
+                flat_grad, wire, t["desc"], t["pack"], t["pack"].shape[0],
+                self.codec.bucket_size, self.codec.qlevel,
+                self.codec.scheme == "terngrad", self._qsgd_seed,
+            )
+            return sum(s.wire_words for s in self.specs)
         grads = [
             p.grad if p.grad is not None else torch.zeros_like(p) for p in self.params
         ]
@@ -177,6 +215,19 @@ class WireCodec:
         """agg (flat, total_params) += sum over rows of ``stacked``."""
         W = stacked.shape[0]
         use_hip = stacked.is_cuda
+        if self._qsgd_tables is not None:
+            from ..ops import ext
+
+            t = self._qsgd_tables
+            for w in range(W):
+                ext().qsgd_unpack_batched(
+                    stacked[w].contiguous()
+                    if not stacked[w].is_contiguous()
+                    else stacked[w],
+                    agg, t["desc"], t["unpack"], t["unpack"].shape[0],
+                    self.codec.bucket_size, self.codec.qlevel,
+                )
+            return
         for spec, p_off in zip(self.specs, self.param_offsets):
             regions = stacked.narrow(1, spec.wire_offset, spec.wire_words)
             out = agg[p_off : p_off + spec.numel]
