@@ -30,7 +30,8 @@ void atomo_batched_sel_launch(const float*, float*, const float*,
                               const int64_t*, const int32_t*, int, int,
                               hipStream_t);
 void atomo_jacobi_eigh_launch(float*, float*, const int64_t*, const int64_t*,
-                              const int32_t*, int, int, hipStream_t);
+                              const int32_t*, int, int, float*,
+                              const int64_t*, int, hipStream_t);
 void atomo_jacobi_eigh_big_launch(float*, float*, float*, const int64_t*,
                                   const int64_t*, const int32_t*,
                                   const int64_t*, int, hipStream_t);
@@ -186,7 +187,8 @@ void batched_sel(torch::Tensor flat, torch::Tensor wire, torch::Tensor stage,
 
 void jacobi_eigh(torch::Tensor grams, torch::Tensor evals, torch::Tensor desc,
                  torch::Tensor eval_offs, torch::Tensor rows, int64_t n_mats,
-                 int64_t jmax) {
+                 int64_t jmax, torch::Tensor vwarm, torch::Tensor vwarm_offs,
+                 bool warm) {
   check_f32_cuda(grams, "grams");
   check_f32_cuda(evals, "evals");
   TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kInt64 &&
@@ -197,11 +199,13 @@ void jacobi_eigh(torch::Tensor grams, torch::Tensor evals, torch::Tensor desc,
   TORCH_CHECK(rows.is_cuda() && rows.scalar_type() == torch::kInt32,
               "rows must be cuda int32");
   if (n_mats == 0) return;
-  atomo_jacobi_eigh_launch(grams.data_ptr<float>(), evals.data_ptr<float>(),
-                           desc.data_ptr<int64_t>(),
-                           eval_offs.data_ptr<int64_t>(),
-                           rows.data_ptr<int32_t>(), (int)n_mats, (int)jmax,
-                           cur_stream());
+  atomo_jacobi_eigh_launch(
+      grams.data_ptr<float>(), evals.data_ptr<float>(),
+      desc.data_ptr<int64_t>(), eval_offs.data_ptr<int64_t>(),
+      rows.data_ptr<int32_t>(), (int)n_mats, (int)jmax,
+      vwarm.numel() ? vwarm.data_ptr<float>() : nullptr,
+      vwarm_offs.numel() ? vwarm_offs.data_ptr<int64_t>() : nullptr,
+      warm ? 1 : 0, cur_stream());
 }
 
 void jacobi_eigh_big(torch::Tensor grams, torch::Tensor vbuf,

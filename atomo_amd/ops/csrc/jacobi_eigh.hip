@@ -52,14 +52,21 @@ __device__ __forceinline__ void rot_params(float app, float aqq, float apq,
 // ---------------------------------------------------------------------------
 // small variant: LDS-resident
 // ---------------------------------------------------------------------------
-template <int JMAX>
+// vwarm: persistent per-layer eigenvector basis from the PREVIOUS step.
+// Gradients (hence Grams) evolve slowly, so pre-rotating B = Vw^T G Vw
+// leaves B nearly diagonal and the sweep loop exits after ~1-2 sweeps
+// instead of ~6-8 from scratch.  The final basis is V = Vw . R and is
+// written back to vwarm for the next step.
+template <int JMAX, bool WARM>
 __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
     float* __restrict__ grams, float* __restrict__ evals,
     const int64_t* __restrict__ desc, const int64_t* __restrict__ eval_offs,
-    const int32_t* __restrict__ rows_list, int n_mats) {
+    const int32_t* __restrict__ rows_list, int n_mats,
+    float* __restrict__ vwarm, const int64_t* __restrict__ vwarm_offs) {
   constexpr int JSTRIDE = JMAX + 1;
   __shared__ float G[JMAX * JSTRIDE];
   __shared__ float V[JMAX * JSTRIDE];
+  __shared__ float T[WARM ? JMAX * JSTRIDE : 1];
   __shared__ float cs[JMAX / 2], sn[JMAX / 2];
   __shared__ int pp[JMAX / 2], qq[JMAX / 2];
   __shared__ int order[JMAX];
@@ -74,6 +81,7 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
   const int sm = is_tall ? n : m;
   float* Gg = grams + d[4];
   float* ev = evals + eval_offs[row];
+  float* Vw = WARM ? vwarm + vwarm_offs[blockIdx.x] : nullptr;
   const int tid = threadIdx.x;
   const int N = (sm + 1) & ~1;  // even-padded
 
@@ -82,7 +90,11 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
     const int r = i / N, c = i % N;
     const float g = (r < sm && c < sm) ? Gg[r * sm + c] : 0.f;
     G[r * JSTRIDE + c] = g;
-    V[r * JSTRIDE + c] = (r == c) ? 1.f : 0.f;
+    if (WARM)
+      V[r * JSTRIDE + c] = (r < sm && c < sm) ? Vw[r * sm + c]
+                                              : ((r == c) ? 1.f : 0.f);
+    else
+      V[r * JSTRIDE + c] = (r == c) ? 1.f : 0.f;
     fro2 += g * g;
   }
 #pragma unroll
@@ -90,6 +102,25 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
   if ((tid & 63) == 0) offsq[tid >> 6] = fro2;
   if (tid == 0) done_s = 0;
   __syncthreads();
+  if (WARM) {
+    // B = V^T G V  (T = G.V, then G = V^T.T), all in LDS
+    for (int i = tid; i < N * N; i += JTHREADS) {
+      const int r = i / N, c = i % N;
+      float acc = 0.f;
+      for (int k = 0; k < N; ++k)
+        acc = fmaf(G[r * JSTRIDE + k], V[k * JSTRIDE + c], acc);
+      T[r * JSTRIDE + c] = acc;
+    }
+    __syncthreads();
+    for (int i = tid; i < N * N; i += JTHREADS) {
+      const int r = i / N, c = i % N;
+      float acc = 0.f;
+      for (int k = 0; k < N; ++k)
+        acc = fmaf(V[k * JSTRIDE + r], T[k * JSTRIDE + c], acc);
+      G[r * JSTRIDE + c] = acc;
+    }
+    __syncthreads();
+  }
   float fro_all = 0.f;
   for (int w = 0; w < JTHREADS / 64; ++w) fro_all += offsq[w];
   const float tol2 = fro_all * 1e-13f;
@@ -173,7 +204,9 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
   __syncthreads();
   for (int i = tid; i < sm * sm; i += JTHREADS) {
     const int k = i / sm, j = i % sm;
-    Gg[i] = V[k * JSTRIDE + order[j]];
+    const float v = V[k * JSTRIDE + order[j]];
+    Gg[i] = v;
+    if (WARM) Vw[i] = v;
   }
 }
 
@@ -495,15 +528,22 @@ void atomo_sample_stage_launch(const float* evecs, const float* evals,
 void atomo_jacobi_eigh_launch(float* grams, float* evals, const int64_t* desc,
                               const int64_t* eval_offs,
                               const int32_t* rows_list, int n_mats, int jmax,
-                              hipStream_t stream) {
-  if (jmax <= 64)
-    hipLaunchKernelGGL(jacobi_eigh_kernel<64>, dim3(n_mats), dim3(JTHREADS),
-                       0, stream, grams, evals, desc, eval_offs, rows_list,
-                       n_mats);
-  else
-    hipLaunchKernelGGL(jacobi_eigh_kernel<128>, dim3(n_mats), dim3(JTHREADS),
-                       0, stream, grams, evals, desc, eval_offs, rows_list,
-                       n_mats);
+                              float* vwarm, const int64_t* vwarm_offs,
+                              int warm, hipStream_t stream) {
+  if (jmax <= 64) {
+    if (warm && vwarm != nullptr)
+      hipLaunchKernelGGL((jacobi_eigh_kernel<64, true>), dim3(n_mats),
+                         dim3(JTHREADS), 0, stream, grams, evals, desc,
+                         eval_offs, rows_list, n_mats, vwarm, vwarm_offs);
+    else
+      hipLaunchKernelGGL((jacobi_eigh_kernel<64, false>), dim3(n_mats),
+                         dim3(JTHREADS), 0, stream, grams, evals, desc,
+                         eval_offs, rows_list, n_mats, vwarm, vwarm_offs);
+  } else {
+    hipLaunchKernelGGL((jacobi_eigh_kernel<128, false>), dim3(n_mats),
+                       dim3(JTHREADS), 0, stream, grams, evals, desc,
+                       eval_offs, rows_list, n_mats, vwarm, vwarm_offs);
+  }
 }
 
 void atomo_jacobi_eigh_big_launch(float* grams, float* vbuf, float* evals,

@@ -183,6 +183,16 @@ class BatchedSVDEncoder:
                 self.rows_j64 = torch.tensor(rows_j64, dtype=torch.int32, device=dev)
                 self.rows_j128 = torch.tensor(rows_j128, dtype=torch.int32, device=dev)
                 self.layer_row = {i: r for r, i in enumerate(self.kernel_rows)}
+                # warm-start basis for the sm<=64 Jacobi (persist V per layer)
+                vw_offs, vw = [], 0
+                row_to_layer = {r: i for i, r in self.layer_row.items()}
+                for r in rows_j64:
+                    sm = self.small[row_to_layer[r]]
+                    vw_offs.append(vw)
+                    vw += sm * sm
+                self.vwarm = torch.zeros(max(1, vw), dtype=torch.float32, device=dev)
+                self.vwarm_offs = torch.tensor(vw_offs, dtype=torch.int64, device=dev)
+                self._warm = False
                 self.eval_offs_dev = torch.tensor(eval_offs, dtype=torch.int64, device=dev)
                 self.eval_offs = eval_offs
                 self.evals_dev = torch.zeros(max(1, ev_off), dtype=torch.float32, device=dev)
@@ -320,12 +330,15 @@ class BatchedSVDEncoder:
                 e.jacobi_eigh(
                     self.grams, self.evals_dev, self.desc, self.eval_offs_dev,
                     self.rows_j64, self.rows_j64.shape[0], 64,
+                    self.vwarm, self.vwarm_offs, self._warm,
                 )
+                self._warm = True
                 if self.rows_j128.shape[0]:
                     e.jacobi_eigh(
                         self.grams, self.evals_dev, self.desc,
                         self.eval_offs_dev, self.rows_j128,
                         self.rows_j128.shape[0], 128,
+                        self.vwarm, self.vwarm_offs, False,
                     )
                 # big folds: batched hipSOLVER syevd per size-group, each
                 # group on its own stream so independent solves overlap
