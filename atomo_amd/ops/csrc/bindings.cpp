@@ -188,7 +188,7 @@ void batched_sel(torch::Tensor flat, torch::Tensor wire, torch::Tensor stage,
 void jacobi_eigh(torch::Tensor grams, torch::Tensor evals, torch::Tensor desc,
                  torch::Tensor eval_offs, torch::Tensor rows, int64_t n_mats,
                  int64_t jmax, torch::Tensor vwarm, torch::Tensor vwarm_offs,
-                 bool warm) {
+                 int64_t warm) {
   check_f32_cuda(grams, "grams");
   check_f32_cuda(evals, "evals");
   TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kInt64 &&
@@ -203,9 +203,9 @@ void jacobi_eigh(torch::Tensor grams, torch::Tensor evals, torch::Tensor desc,
       grams.data_ptr<float>(), evals.data_ptr<float>(),
       desc.data_ptr<int64_t>(), eval_offs.data_ptr<int64_t>(),
       rows.data_ptr<int32_t>(), (int)n_mats, (int)jmax,
-      vwarm.numel() ? vwarm.data_ptr<float>() : nullptr,
+      vwarm.numel() > 1 ? vwarm.data_ptr<float>() : nullptr,
       vwarm_offs.numel() ? vwarm_offs.data_ptr<int64_t>() : nullptr,
-      warm ? 1 : 0, cur_stream());
+      (int)warm, cur_stream());
 }
 
 void jacobi_eigh_big(torch::Tensor grams, torch::Tensor vbuf,

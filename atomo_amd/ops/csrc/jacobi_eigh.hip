@@ -62,7 +62,8 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
     float* __restrict__ grams, float* __restrict__ evals,
     const int64_t* __restrict__ desc, const int64_t* __restrict__ eval_offs,
     const int32_t* __restrict__ rows_list, int n_mats,
-    float* __restrict__ vwarm, const int64_t* __restrict__ vwarm_offs) {
+    float* __restrict__ vwarm, const int64_t* __restrict__ vwarm_offs,
+    int save_warm) {
   constexpr int JSTRIDE = JMAX + 1;
   __shared__ float G[JMAX * JSTRIDE];
   __shared__ float V[JMAX * JSTRIDE];
@@ -81,7 +82,7 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
   const int sm = is_tall ? n : m;
   float* Gg = grams + d[4];
   float* ev = evals + eval_offs[row];
-  float* Vw = WARM ? vwarm + vwarm_offs[blockIdx.x] : nullptr;
+  float* Vw = (WARM || save_warm) ? vwarm + vwarm_offs[blockIdx.x] : nullptr;
   const int tid = threadIdx.x;
   const int N = (sm + 1) & ~1;  // even-padded
 
@@ -206,7 +207,7 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
     const int k = i / sm, j = i % sm;
     const float v = V[k * JSTRIDE + order[j]];
     Gg[i] = v;
-    if (WARM) Vw[i] = v;
+    if (WARM || save_warm) Vw[i] = v;
   }
 }
 
@@ -530,19 +531,23 @@ void atomo_jacobi_eigh_launch(float* grams, float* evals, const int64_t* desc,
                               const int32_t* rows_list, int n_mats, int jmax,
                               float* vwarm, const int64_t* vwarm_offs,
                               int warm, hipStream_t stream) {
+  // warm < 0: no warm basis at all (also skip the save)
+  const int save = (warm >= 0 && vwarm != nullptr) ? 1 : 0;
   if (jmax <= 64) {
-    if (warm && vwarm != nullptr)
+    if (warm > 0 && vwarm != nullptr)
       hipLaunchKernelGGL((jacobi_eigh_kernel<64, true>), dim3(n_mats),
                          dim3(JTHREADS), 0, stream, grams, evals, desc,
-                         eval_offs, rows_list, n_mats, vwarm, vwarm_offs);
+                         eval_offs, rows_list, n_mats, vwarm, vwarm_offs,
+                         save);
     else
       hipLaunchKernelGGL((jacobi_eigh_kernel<64, false>), dim3(n_mats),
                          dim3(JTHREADS), 0, stream, grams, evals, desc,
-                         eval_offs, rows_list, n_mats, vwarm, vwarm_offs);
+                         eval_offs, rows_list, n_mats, vwarm, vwarm_offs,
+                         save);
   } else {
     hipLaunchKernelGGL((jacobi_eigh_kernel<128, false>), dim3(n_mats),
                        dim3(JTHREADS), 0, stream, grams, evals, desc,
-                       eval_offs, rows_list, n_mats, vwarm, vwarm_offs);
+                       eval_offs, rows_list, n_mats, vwarm, vwarm_offs, 0);
   }
 }
 

@@ -330,7 +330,7 @@ class BatchedSVDEncoder:
                 e.jacobi_eigh(
                     self.grams, self.evals_dev, self.desc, self.eval_offs_dev,
                     self.rows_j64, self.rows_j64.shape[0], 64,
-                    self.vwarm, self.vwarm_offs, self._warm,
+                    self.vwarm, self.vwarm_offs, 1 if self._warm else 0,
                 )
                 self._warm = True
                 if self.rows_j128.shape[0]:
@@ -338,7 +338,7 @@ class BatchedSVDEncoder:
                         self.grams, self.evals_dev, self.desc,
                         self.eval_offs_dev, self.rows_j128,
                         self.rows_j128.shape[0], 128,
-                        self.vwarm, self.vwarm_offs, False,
+                        self.vwarm, self.vwarm_offs, -1,
                     )
                 # big folds: batched hipSOLVER syevd per size-group, each
                 # group on its own stream so independent solves overlap

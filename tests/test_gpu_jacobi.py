@@ -28,7 +28,7 @@ def _run_jacobi(sm, B=5, seed=None):
         vwarm = torch.zeros(1, device=dev)
         vw_offs = torch.zeros(B, dtype=torch.int64, device=dev)
         ext().jacobi_eigh(grams, evals, desc, eval_offs, rows, B,
-                          64 if sm <= 64 else 128, vwarm, vw_offs, False)
+                          64 if sm <= 64 else 128, vwarm, vw_offs, -1)
     else:
         v_offs = torch.tensor(
             [b * sm * sm for b in range(B)], dtype=torch.int64, device=dev
