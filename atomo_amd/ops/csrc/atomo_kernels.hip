@@ -339,6 +339,51 @@ __global__ void svd_decode_acc_kernel(const float* __restrict__ regions,
 }
 
 // ---------------------------------------------------------------------------
+// Batched SVD decode over ALL layers and ALL workers in one launch.
+// Descriptor row (int64, DD_N words): [agg_off, m, n, r_max, wire_off]
+// Work map: (layer, chunk of DEC_CHUNK output elements).
+// agg[agg_off + t] += sum_w sum_r u_w[i,r] s_w[r] vT_w[r,j], t=(i,j)
+// ---------------------------------------------------------------------------
+#define DD_N 5
+#define DEC_CHUNK 1024
+
+__global__ void svd_decode_batched_kernel(
+    const float* __restrict__ stacked, int64_t row_stride, int W,
+    float* __restrict__ agg, const int64_t* __restrict__ desc,
+    const int32_t* __restrict__ work, int n_tiles) {
+  for (int tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+    const int layer = work[2 * tile];
+    const int chunk = work[2 * tile + 1];
+    const int64_t* d = desc + (int64_t)layer * DD_N;
+    const int64_t agg_off = d[0];
+    const int m = (int)d[1], n = (int)d[2];
+    const int r_max = (int)d[3];
+    const int64_t wo = d[4];
+    const int64_t total = (int64_t)m * n;
+    const int64_t s_off = wo + 1 + (int64_t)r_max * m;
+    const int64_t v_off = s_off + r_max;
+    const int64_t t0 = (int64_t)chunk * DEC_CHUNK;
+    const int64_t t1 = min(t0 + DEC_CHUNK, total);
+    for (int64_t t = t0 + threadIdx.x; t < t1; t += blockDim.x) {
+      const int i = (int)(t / n);
+      const int j = (int)(t % n);
+      float acc = 0.f;
+      for (int w = 0; w < W; ++w) {
+        const float* row = stacked + (int64_t)w * row_stride;
+        const int r_hat = (int)row[wo];
+        const float* uT = row + wo + 1;
+        const float* s = row + s_off;
+        const float* vT = row + v_off;
+        for (int r = 0; r < r_hat; ++r)
+          acc = fmaf(uT[(int64_t)r * m + i] * s[r], vT[(int64_t)r * n + j],
+                     acc);
+      }
+      agg[agg_off + t] += acc;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Fused SGD: p -= lr * step(grad*scale + wd*p, momentum buffer)
 // ---------------------------------------------------------------------------
 __global__ void fused_sgd_kernel(float* __restrict__ p,
@@ -424,6 +469,15 @@ void atomo_qsgd_unpack_batched_launch(const float* wire, float* agg,
   hipLaunchKernelGGL(qsgd_unpack_batched_kernel, dim3(grid), dim3(256), 0,
                      stream, wire, agg, desc, work, n_tiles, bucket_size,
                      qlevel);
+}
+
+void atomo_svd_decode_batched_launch(const float* stacked, int64_t row_stride,
+                                     int W, float* agg, const int64_t* desc,
+                                     const int32_t* work, int n_tiles,
+                                     hipStream_t stream) {
+  const int grid = n_tiles > 8192 ? 8192 : (n_tiles < 1 ? 1 : n_tiles);
+  hipLaunchKernelGGL(svd_decode_batched_kernel, dim3(grid), dim3(256), 0,
+                     stream, stacked, row_stride, W, agg, desc, work, n_tiles);
 }
 
 void atomo_fused_sgd_launch(float* p, const float* g, float* buf, int64_t n,

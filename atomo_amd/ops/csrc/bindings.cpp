@@ -18,6 +18,9 @@ void atomo_svd_decode_acc_launch(const float*, float*, int, int64_t, int, int,
                                  int, hipStream_t);
 void atomo_fused_sgd_launch(float*, const float*, float*, int64_t, float,
                             float, float, bool, float, float, hipStream_t);
+void atomo_svd_decode_batched_launch(const float*, int64_t, int, float*,
+                                     const int64_t*, const int32_t*, int,
+                                     hipStream_t);
 void atomo_qsgd_pack_batched_launch(const float*, float*, const int64_t*,
                                     const int32_t*, int, int, int, bool,
                                     uint64_t, hipStream_t);
@@ -122,6 +125,20 @@ void fused_sgd(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
                          (float)lr, (float)momentum, (float)weight_decay,
                          nesterov, (float)dampening, (float)grad_scale,
                          cur_stream());
+}
+
+void svd_decode_batched(torch::Tensor stacked, torch::Tensor agg,
+                        torch::Tensor desc, torch::Tensor work,
+                        int64_t n_tiles) {
+  TORCH_CHECK(stacked.is_cuda() && stacked.scalar_type() == torch::kFloat32 &&
+                  stacked.dim() == 2 && stacked.stride(1) == 1,
+              "stacked must be cuda fp32 (W, words) row-contiguous");
+  check_f32_cuda(agg, "agg");
+  if (n_tiles == 0) return;
+  atomo_svd_decode_batched_launch(
+      stacked.data_ptr<float>(), stacked.stride(0), (int)stacked.size(0),
+      agg.data_ptr<float>(), desc.data_ptr<int64_t>(),
+      work.data_ptr<int32_t>(), (int)n_tiles, cur_stream());
 }
 
 void qsgd_pack_batched(torch::Tensor flat, torch::Tensor wire,
@@ -274,6 +291,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("batched_sel", &batched_sel,
         "batched selection GEMM + packet scatter into the wire");
   m.def("qsgd_pack", &qsgd_pack, "QSGD bucket quantize+pack (gfx950)");
+  m.def("svd_decode_batched", &svd_decode_batched,
+        "one-launch SVD decode+accumulate over all layers and workers");
   m.def("qsgd_pack_batched", &qsgd_pack_batched,
         "one-launch QSGD pack over all layers");
   m.def("qsgd_unpack_batched", &qsgd_unpack_batched,

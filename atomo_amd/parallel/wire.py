@@ -81,6 +81,36 @@ class WireCodec:
                     self._pad_scratch[s.index] = torch.zeros(
                         s.meta["padded"], dtype=torch.float32, device=device
                     )
+        # batched SVD decode tables: one launch over all layers x workers
+        self._svd_decode_tables = None
+        if (
+            device.type == "cuda"
+            and isinstance(codec, SVDCodec)
+            and codec.compress
+        ):
+            from .. import ops
+
+            if ops.have_ext():
+                DEC_CHUNK = 1024
+                desc, work, kernel_layers = [], [], set()
+                for s, p_off in zip(self.specs, self.param_offsets):
+                    if s.meta["padded"] != s.numel:
+                        continue  # odd folds decode through the scratch path
+                    meta = s.meta
+                    li = len(desc)
+                    desc.append(
+                        [p_off, meta["m"], meta["n"], meta["r_max"], s.wire_offset]
+                    )
+                    kernel_layers.add(s.index)
+                    total = meta["m"] * meta["n"]
+                    for c in range((total + DEC_CHUNK - 1) // DEC_CHUNK):
+                        work.append([li, c])
+                if desc:
+                    self._svd_decode_tables = {
+                        "desc": torch.tensor(desc, dtype=torch.int64, device=device),
+                        "work": torch.tensor(work, dtype=torch.int32, device=device),
+                        "layers": kernel_layers,
+                    }
 
     @property
     def reducible(self) -> bool:
@@ -228,7 +258,18 @@ class WireCodec:
                     self.codec.bucket_size, self.codec.qlevel,
                 )
             return
+        dec_tables = self._svd_decode_tables if use_hip else None
+        if dec_tables is not None:
+            from ..ops import ext
+
+            ext().svd_decode_batched(
+                stacked, agg, dec_tables["desc"], dec_tables["work"],
+                dec_tables["work"].shape[0],
+            )
+        covered = dec_tables["layers"] if dec_tables is not None else set()
         for spec, p_off in zip(self.specs, self.param_offsets):
+            if spec.index in covered:
+                continue
             regions = stacked.narrow(1, spec.wire_offset, spec.wire_words)
             out = agg[p_off : p_off + spec.numel]
             if isinstance(self.codec, SVDCodec) and self.codec.compress:
