@@ -95,3 +95,44 @@ def test_batched_kernels_unbiased_sampling(dev):
     for j, (g, spec) in enumerate(zip(grads, specs)):
         rel = ((acc[j] / n) - g.reshape(-1)).norm() / g.norm()
         assert rel < 0.5, (spec.shape, rel)
+
+
+def test_batched_decode_matches_per_layer(dev):
+    """svd_decode_batched == per-layer svd_decode_acc on identical packets."""
+    from atomo_amd.codings import SVDCodec
+    from atomo_amd.parallel.wire import WireCodec
+    from atomo_amd.ops import svd_ops
+
+    torch.manual_seed(7)
+    shapes = [(20, 1, 5, 5), (10,), (500, 50), (10, 500), (64, 16, 3, 3)]
+    codec = SVDCodec(rank=3, generator=torch.Generator().manual_seed(3))
+    numels = [int(torch.Size(s).numel()) for s in shapes]
+    params = []
+    for s in shapes:
+        p = torch.nn.Parameter(torch.zeros(s, device=dev))
+        p.grad = torch.randn(s, device=dev)
+        params.append(p)
+    wc = WireCodec(codec, params, dev)
+    assert wc._svd_decode_tables is not None
+    wire = torch.zeros(wc.total_words, device=dev)
+    wc.encode_all(wire)  # per-layer oracle encode (no flat_grad)
+    stacked = torch.stack([wire, wire * 0.0 + wire])  # two identical workers
+
+    agg_new = torch.zeros(sum(numels), device=dev)
+    wc.decode_all(stacked, agg_new)
+
+    agg_ref = torch.zeros(sum(numels), device=dev)
+    for spec, p_off in zip(wc.specs, wc.param_offsets):
+        meta = spec.meta
+        regions = stacked.narrow(1, spec.wire_offset, spec.wire_words)
+        out = agg_ref[p_off : p_off + spec.numel]
+        if meta["padded"] == spec.numel:
+            svd_ops.decode_acc(
+                regions, out.view(meta["m"], meta["n"]), meta["m"], meta["n"],
+                meta["r_max"],
+            )
+        else:
+            for w in range(2):
+                codec.decode_from(regions[w], out, spec)
+    err = (agg_new - agg_ref).abs().max().item()
+    assert err < 1e-4, err
