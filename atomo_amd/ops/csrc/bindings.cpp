@@ -18,6 +18,9 @@ void atomo_svd_decode_acc_launch(const float*, float*, int, int64_t, int, int,
                                  int, hipStream_t);
 void atomo_fused_sgd_launch(float*, const float*, float*, int64_t, float,
                             float, float, bool, float, float, hipStream_t);
+void atomo_fused_adam_launch(float*, const float*, float*, float*, float*,
+                             int64_t, float, float, float, float, float,
+                             float, float, float, hipStream_t);
 void atomo_svd_decode_batched_launch(const float*, int64_t, int, float*,
                                      const int64_t*, const int32_t*, int,
                                      hipStream_t);
@@ -125,6 +128,27 @@ void fused_sgd(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
                          (float)lr, (float)momentum, (float)weight_decay,
                          nesterov, (float)dampening, (float)grad_scale,
                          cur_stream());
+}
+
+void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor exp_avg,
+                torch::Tensor exp_avg_sq, torch::Tensor max_exp_avg_sq,
+                double lr, double beta1, double beta2, double eps,
+                double weight_decay, double bias1, double bias2,
+                double grad_scale) {
+  check_f32_cuda(p, "p");
+  check_f32_cuda(g, "g");
+  check_f32_cuda(exp_avg, "exp_avg");
+  check_f32_cuda(exp_avg_sq, "exp_avg_sq");
+  const int64_t n = p.numel();
+  TORCH_CHECK(g.numel() == n && exp_avg.numel() == n &&
+                  exp_avg_sq.numel() == n,
+              "adam buffer size mismatch");
+  atomo_fused_adam_launch(
+      p.data_ptr<float>(), g.data_ptr<float>(), exp_avg.data_ptr<float>(),
+      exp_avg_sq.data_ptr<float>(),
+      max_exp_avg_sq.numel() ? max_exp_avg_sq.data_ptr<float>() : nullptr, n,
+      (float)lr, (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+      (float)bias1, (float)bias2, (float)grad_scale, cur_stream());
 }
 
 void svd_decode_batched(torch::Tensor stacked, torch::Tensor agg,
@@ -301,4 +325,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("svd_decode_acc", &svd_decode_acc,
         "fused rank-k SVD decode+accumulate over W packets");
   m.def("fused_sgd", &fused_sgd, "fused flat SGD apply");
+  m.def("fused_adam", &fused_adam, "fused flat Adam apply");
 }

@@ -31,10 +31,24 @@ class ExternalAdam:
         self.max_exp_avg_sq = torch.zeros_like(flat_params) if amsgrad else None
 
     @torch.no_grad()
-    def step(self, flat_grad: torch.Tensor) -> None:
+    def step(self, flat_grad: torch.Tensor, grad_scale: float = 1.0) -> None:
         self.t += 1
         b1, b2 = self.betas
+        if self.p.is_cuda:
+            from ..ops import ext
+
+            ext().fused_adam(
+                self.p, flat_grad, self.exp_avg, self.exp_avg_sq,
+                self.max_exp_avg_sq
+                if self.max_exp_avg_sq is not None
+                else torch.empty(0, device=self.p.device),
+                self.lr, b1, b2, self.eps, self.weight_decay,
+                1 - b1 ** self.t, 1 - b2 ** self.t, grad_scale,
+            )
+            return
         g = flat_grad
+        if grad_scale != 1.0:
+            g = g * grad_scale
         if self.weight_decay != 0.0:
             g = g.add(self.p, alpha=self.weight_decay)
         self.exp_avg.mul_(b1).add_(g, alpha=1 - b1)

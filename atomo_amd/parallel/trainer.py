@@ -299,6 +299,9 @@ class PSTrainer:
                 dampening=self.opt.dampening,
                 grad_scale=scale,
             )
+        elif type(self.opt).__name__ == "ExternalAdam":
+            self.opt.lr = self.lr
+            self.opt.step(grad_flat, grad_scale=scale)
         else:
             if scale != 1.0:
                 grad_flat = grad_flat * scale

@@ -188,3 +188,45 @@ def test_fused_sgd_grad_scale(dev):
     buf = torch.zeros(n, device=dev)
     optim_ops.fused_sgd(p, g, buf, lr=1.0, momentum=0.0, grad_scale=0.25)
     assert torch.allclose(p, torch.full_like(p, -1.0))
+
+
+@pytest.mark.parametrize("amsgrad", [False, True])
+def test_fused_adam_matches_cpu(dev, amsgrad):
+    from atomo_amd.optim import ExternalAdam
+
+    torch.manual_seed(8)
+    n = 50001
+    p_cpu = torch.randn(n)
+    p_gpu = p_cpu.to(dev)
+    opt_cpu = ExternalAdam(p_cpu, lr=0.01, weight_decay=1e-4, amsgrad=amsgrad)
+    opt_gpu = ExternalAdam(p_gpu, lr=0.01, weight_decay=1e-4, amsgrad=amsgrad)
+    for _ in range(5):
+        g = torch.randn(n)
+        opt_cpu.step(g, grad_scale=0.5)
+        opt_gpu.step(g.to(dev), grad_scale=0.5)
+    assert torch.allclose(p_gpu.cpu(), p_cpu, atol=1e-5), (
+        (p_gpu.cpu() - p_cpu).abs().max()
+    )
+
+
+def test_qsgd_terngrad_batched_gpu(dev):
+    """Batched terngrad path on GPU: ternary output, bounded by clipped max."""
+    from atomo_amd.codings import QSGDCodec
+    from atomo_amd.parallel.wire import WireCodec
+
+    torch.manual_seed(9)
+    codec = QSGDCodec(quantization_level=1, bucket_size=256, scheme="terngrad")
+    p = torch.nn.Parameter(torch.zeros(512, device=dev))
+    flat = torch.randn(512, device=dev)
+    p.grad = flat.view(512)
+    wc = WireCodec(codec, [p], dev)
+    wire = torch.zeros(wc.total_words, device=dev)
+    wc.encode_all(wire, flat_grad=flat)
+    agg = torch.zeros(512, device=dev)
+    wc.decode_all(wire.view(1, -1), agg)
+    for b in range(2):
+        vals = agg[b * 256 : (b + 1) * 256]
+        mx = vals.abs().max()
+        uniq = vals.unique()
+        for v in uniq:
+            assert torch.isclose(v.abs(), mx, atol=1e-6) or v == 0
