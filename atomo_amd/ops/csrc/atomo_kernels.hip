@@ -405,12 +405,56 @@ __global__ void fused_sgd_kernel(float* __restrict__ p,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fused Adam: one flat sweep (reference math optim/adam.py:37-93, incl.
+// amsgrad), with the PS's 1/num_workers fold into grad_scale.
+// ---------------------------------------------------------------------------
+__global__ void fused_adam_kernel(float* __restrict__ p,
+                                  const float* __restrict__ g,
+                                  float* __restrict__ exp_avg,
+                                  float* __restrict__ exp_avg_sq,
+                                  float* __restrict__ max_exp_avg_sq,
+                                  int64_t n, float lr, float beta1,
+                                  float beta2, float eps, float weight_decay,
+                                  float bias1, float bias2, float grad_scale) {
+  const float step_size = lr / bias1;
+  const float inv_bias2 = 1.0f / bias2;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const float pi = p[i];
+    const float gi = g[i] * grad_scale + weight_decay * pi;
+    const float m = exp_avg[i] * beta1 + (1.f - beta1) * gi;
+    float v = exp_avg_sq[i] * beta2 + (1.f - beta2) * gi * gi;
+    exp_avg[i] = m;
+    exp_avg_sq[i] = v;
+    if (max_exp_avg_sq != nullptr) {
+      v = fmaxf(max_exp_avg_sq[i], v);
+      max_exp_avg_sq[i] = v;
+    }
+    const float denom = sqrtf(v * inv_bias2) + eps;
+    p[i] = pi - step_size * m / denom;
+  }
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
 // extern "C" launchers (bindings.cpp provides the torch glue)
 // ---------------------------------------------------------------------------
 extern "C" {
+
+void atomo_fused_adam_launch(float* p, const float* g, float* exp_avg,
+                             float* exp_avg_sq, float* max_exp_avg_sq,
+                             int64_t n, float lr, float beta1, float beta2,
+                             float eps, float weight_decay, float bias1,
+                             float bias2, float grad_scale,
+                             hipStream_t stream) {
+  const int block = 256;
+  const int grid = grid_for(n, block);
+  hipLaunchKernelGGL(fused_adam_kernel, dim3(grid), dim3(block), 0, stream, p,
+                     g, exp_avg, exp_avg_sq, max_exp_avg_sq, n, lr, beta1,
+                     beta2, eps, weight_decay, bias1, bias2, grad_scale);
+}
 
 void atomo_qsgd_pack_launch(const float* grad, float* norms, uint32_t* packed,
                             int64_t numel, int bucket_size, int qlevel,
