@@ -217,31 +217,38 @@ class BatchedSVDEncoder:
             return scratch.view(m, n)
         return flat.view(m, n)
 
-    def _sample_all(self, svals_h: dict) -> dict:
-        """Vectorized Bernoulli sampling over every layer (reference
-        _sample_svd semantics, svd.py:49-67).  svals_h: layer -> fp32/fp64
-        singular values (descending).  Returns layer -> (idx, probs)."""
+    def _sample_all(self, svals_h: dict, layers=None) -> dict:
+        """Vectorized Bernoulli sampling (reference _sample_svd semantics,
+        svd.py:49-67) over ``layers`` (default: all).  svals_h: layer ->
+        fp32/fp64 singular values (descending).  Returns layer ->
+        (idx, probs)."""
+        layers = list(range(len(self.specs))) if layers is None else list(layers)
         samples = {}
         if not self.codec.random_sample:
-            for i, spec in enumerate(self.specs):
+            for i in layers:
+                spec = self.specs[i]
                 r_max = spec.meta["r_max"]
                 r = min(self.codec.rank, r_max) if self.codec.rank > 0 else r_max
                 samples[i] = (torch.arange(r), None)
             return samples
         rank = self.codec.rank
-        probs_list = []
-        for i in range(len(self.specs)):
+        probs_list = {}
+        for i in layers:
             s = svals_h[i].float()
             if s.numel() == 0 or float(s[0]) < 1e-6:
-                probs_list.append(torch.zeros(0))
+                probs_list[i] = torch.zeros(0)
                 samples[i] = (torch.tensor([0]), torch.tensor([1.0]))
                 continue
             p = (s / s[0]) if rank == 0 else (rank * s / s.sum())
-            probs_list.append(p.clamp(max=1.0))
-        cat = torch.cat(probs_list) if probs_list else torch.zeros(0)
+            probs_list[i] = p.clamp(max=1.0)
+        cat = (
+            torch.cat([probs_list[i] for i in layers])
+            if layers
+            else torch.zeros(0)
+        )
         draws = torch.rand(cat.shape, generator=self.codec.generator) < cat
         off = 0
-        for i in range(len(self.specs)):
+        for i in layers:
             p = probs_list[i]
             k = p.numel()
             d = draws[off : off + k]
@@ -257,7 +264,8 @@ class BatchedSVDEncoder:
             else:
                 samples[i] = (idx, p[idx])
         # wire-budget cap: keep the highest-probability atoms
-        for i, spec in enumerate(self.specs):
+        for i in layers:
+            spec = self.specs[i]
             idx, pr = samples[i]
             r_max = spec.meta["r_max"]
             if idx.numel() > r_max:
@@ -385,9 +393,13 @@ class BatchedSVDEncoder:
         else:
             grams_host = self.grams
 
-        # ---- fully-async path: on-device sampler, no host sync ---------
-        # (every layer on device, RNG not pinned to a host generator)
-        if use_kernels and not host_layers and self.codec.generator is None:
+        # ---- on-device sampler for every device layer ------------------
+        # (RNG not pinned to a host generator); host layers, if any, go
+        # through the LAPACK + host-sampling flow below and the two merge
+        # in the wire.  With no host layers the step has ZERO host syncs.
+        device_sampled = use_kernels and self.codec.generator is None
+        self.device_counted = device_sampled
+        if device_sampled:
             self._seed = (self._seed * 6364136223846793005 + 1442695040888963407) % (
                 1 << 62
             )
@@ -401,8 +413,9 @@ class BatchedSVDEncoder:
                 flat_grad, wire, self.stage_dev, self.desc, self.sel_work,
                 self.sel_work.shape[0], self.sel_elems,
             )
-            mark("async sample+sel")
-            return -1  # used words accumulate in used_words_dev
+            if not host_layers:
+                mark("async sample+sel")
+                return -1  # used words accumulate in used_words_dev
 
         def _solve_group(item):
             sm, idxs = item
@@ -447,7 +460,7 @@ class BatchedSVDEncoder:
         if self.device.type == "cuda" and use_kernels:
             torch.cuda.synchronize()  # waits for Jacobi + evals D2H
         mark("B d2h")
-        if use_kernels:
+        if use_kernels and not device_sampled:
             for row, i in enumerate(self.kernel_rows):
                 sm = self.small[i]
                 o = self.eval_offs[row]
@@ -460,14 +473,18 @@ class BatchedSVDEncoder:
                     evecs_h[i] = evecs[j]
         mark("B eigh")
 
-        samples = self._sample_all(svals_h)
+        sample_layers = host_layers if device_sampled else None
+        samples = self._sample_all(svals_h, layers=sample_layers)
         used = 0
-        for i, spec in enumerate(specs):
+        for i in (
+            host_layers if device_sampled else range(len(specs))
+        ):
+            spec = specs[i]
             idx, _ = samples[i]
             used += 1 + idx.numel() * (spec.meta["m"] + spec.meta["n"] + 1)
 
-        # device layers: fill the selection table
-        if use_kernels:
+        # device layers: fill the selection table (host-sampled mode only)
+        if use_kernels and not device_sampled:
             st = self.sel_table_host
             st.zero_()
             for row, i in enumerate(self.kernel_rows):
@@ -515,7 +532,7 @@ class BatchedSVDEncoder:
         mark("B sample+stage")
 
         # ---- phase C: device stage build + batched sel + rocBLAS -------
-        if use_kernels:
+        if use_kernels and not device_sampled:
             e.build_stage(
                 self.grams, self.evals_dev, self.sel_table_dev, self.stage_dev,
                 self.desc, self.eval_offs_dev, len(self.kernel_rows),

@@ -231,7 +231,7 @@ class WireCodec:
             used = self._batched_encoder.encode_all(
                 grads, wire, flat_grad=flat_grad, grams_done=overlap_done
             )
-            if used < 0:
+            if getattr(self._batched_encoder, "device_counted", False):
                 self._device_counted = True
             return used
         used = 0

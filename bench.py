@@ -128,12 +128,12 @@ def main():
     images_per_sec = iters_per_sec * a.batch_size * n_workers
     msg_bytes = trainer.timers.scalars.get("msg_bytes", 0.0)
     steps_counted = max(1, trainer.timers.counts.get("msg_bytes", 1))
+    # device counter covers warmup + timed steps; the host scalar covers the
+    # timed steps of host-path layers — the two partition the layers
     dev_bytes = trainer.wc.device_msg_bytes()
-    if dev_bytes > 0:
-        # device-side counter covers warmup + timed steps
-        grad_mb_per_step = dev_bytes / (a.steps + a.warmup) / 1e6
-    else:
-        grad_mb_per_step = (msg_bytes / steps_counted) / 1e6
+    grad_mb_per_step = (
+        msg_bytes / steps_counted + dev_bytes / (a.steps + a.warmup)
+    ) / 1e6
 
     if comm.rank == 0:
         print(
