@@ -123,3 +123,28 @@ def test_msg_bytes_counted():
     msg = trainer.timers.scalars["msg_bytes"]
     total_grad_bytes = 4 * trainer.flat.numel()
     assert 0 < msg < total_grad_bytes  # compression actually compresses
+
+
+def test_adam_optimizer_in_trainer():
+    comm = Comm(device=torch.device("cpu"))
+    codec = make_codec("sgd")
+    trainer = PSTrainer(
+        model_name="LeNet", codec=codec, comm=comm, lr=0.003,
+        optimizer="adam", num_classes=10, in_channels=1, seed=7,
+        device=torch.device("cpu"),
+    )
+    train, _ = make_loaders("mnist", 16, 16, torch.device("cpu"), seed=9)
+    losses = [trainer.train_step(x, y) for _, (x, y) in zip(range(12), train)]
+    assert losses[-1] < losses[0]
+
+
+def test_compress_false_svd_behaves_like_raw():
+    comm = Comm(device=torch.device("cpu"))
+    codec = make_codec("svd", rank=3, compress=False)
+    trainer = PSTrainer(
+        model_name="LeNet", codec=codec, comm=comm, lr=0.05,
+        num_classes=10, in_channels=1, seed=7, device=torch.device("cpu"),
+    )
+    train, _ = make_loaders("mnist", 16, 16, torch.device("cpu"), seed=9)
+    losses = [trainer.train_step(x, y) for _, (x, y) in zip(range(8), train)]
+    assert losses[-1] < losses[0]
