@@ -134,8 +134,8 @@ def test_adam_optimizer_in_trainer():
         device=torch.device("cpu"),
     )
     train, _ = make_loaders("mnist", 16, 16, torch.device("cpu"), seed=9)
-    losses = [trainer.train_step(x, y) for _, (x, y) in zip(range(12), train)]
-    assert losses[-1] < losses[0]
+    losses = [trainer.train_step(x, y) for _, (x, y) in zip(range(30), train)]
+    assert sum(losses[-5:]) / 5 < sum(losses[:5]) / 5
 
 
 def test_compress_false_svd_behaves_like_raw():
@@ -146,5 +146,5 @@ def test_compress_false_svd_behaves_like_raw():
         num_classes=10, in_channels=1, seed=7, device=torch.device("cpu"),
     )
     train, _ = make_loaders("mnist", 16, 16, torch.device("cpu"), seed=9)
-    losses = [trainer.train_step(x, y) for _, (x, y) in zip(range(8), train)]
-    assert losses[-1] < losses[0]
+    losses = [trainer.train_step(x, y) for _, (x, y) in zip(range(20), train)]
+    assert sum(losses[-5:]) / 5 < sum(losses[:5]) / 5
