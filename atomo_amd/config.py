@@ -70,6 +70,9 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
                    help="encode layers on a side stream as backward produces them")
     p.add_argument("--graph", action="store_true", default=False,
                    help="capture forward/backward in a hipGraph and replay")
+    p.add_argument("--amp", action="store_true", default=False,
+                   help="bf16 autocast for forward/backward (fp32 grads, "
+                        "codec and optimizer unchanged)")
     p.add_argument("--step-timeout", type=float, default=0.0,
                    help="abort the rank if a global step stalls this many "
                         "seconds (straggler/hang watchdog; 0 = off)")
@@ -122,6 +125,7 @@ class RunConfig:
             use_graph=a.graph,
             overlap=a.overlap,
             step_timeout=a.step_timeout,
+            amp=a.amp,
         )
 
 

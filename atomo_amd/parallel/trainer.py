@@ -62,6 +62,7 @@ class PSTrainer:
         overlap: bool = False,
         step_timeout: float = 0.0,
         defer_loss: bool = False,
+        amp: bool = False,
     ):
         self.comm = comm
         self.device = device or comm.device
@@ -134,6 +135,10 @@ class PSTrainer:
         # it via current_loss() when logging.
         self.defer_loss = bool(defer_loss)
         self._loss_tensor = None
+        # opt-in bf16 autocast for forward/backward (gradients and the whole
+        # codec/optimizer path stay fp32; NOT used by the default benchmark,
+        # whose dtype matches the reference's fp32)
+        self.amp = bool(amp) and self.device.type == "cuda"
         self.watchdog = None
         if step_timeout and step_timeout > 0:
             from ..utils.watchdog import StepWatchdog
@@ -181,8 +186,13 @@ class PSTrainer:
                     self._fwd_bwd_graphed(x, y)
                 else:
                     self.flat_grad.zero_()
-                    out = self.model(x)
-                    loss = self.loss_fn(out, y)
+                    if self.amp:
+                        with torch.autocast("cuda", dtype=torch.bfloat16):
+                            out = self.model(x)
+                            loss = self.loss_fn(out, y)
+                    else:
+                        out = self.model(x)
+                        loss = self.loss_fn(out, y)
                     loss.backward()
                     if self.defer_loss:
                         self._loss_tensor = loss.detach()

@@ -46,6 +46,8 @@ def main():
                    help="NHWC memory format for convs")
     p.add_argument("--overlap", action="store_true", default=False,
                    help="backward-hook per-layer encode on a side stream")
+    p.add_argument("--amp", action="store_true", default=False,
+                   help="bf16 autocast fwd/bwd (reported dtype changes)")
     a = p.parse_args()
 
     from atomo_amd.codings import make_codec
@@ -80,6 +82,7 @@ def main():
         use_graph=a.graph,
         overlap=a.overlap,
         defer_loss=True,
+        amp=a.amp,
     )
     if a.channels_last:
         trainer.model.to(memory_format=torch.channels_last)
@@ -149,7 +152,7 @@ def main():
                     "higher_is_better": True,
                     "scaling": "weak",
                     "vs_baseline": None,
-                    "dtype": "fp32",
+                    "dtype": "bf16-autocast" if a.amp else "fp32",
                     "data": "synthetic (CIFAR-10-shaped random, random-init weights)",
                     "iters_per_sec": iters_per_sec,
                     "grad_mb_per_step_per_worker": grad_mb_per_step,
