@@ -72,6 +72,19 @@ class BatchedSVDEncoder:
         self.specs = list(specs)
         self.param_offsets = param_offsets
         self._pool = ThreadPoolExecutor(max_workers=8)
+        # MKL builds a fresh OpenMP team the first time a NEW thread calls
+        # LAPACK (~20 ms each); warm every pool worker once so in-step host
+        # eigensolves run at their true sub-ms cost.
+        if device.type == "cuda":
+            def _mkl_warm():
+                g32 = torch.eye(100) + 0.01 * torch.randn(100, 100)
+                torch.linalg.eigh(g32 @ g32.t())
+                g64 = (g32 @ g32.t()).to(torch.float64)
+                torch.linalg.eigh(g64)
+                return True
+
+            for f in [self._pool.submit(_mkl_warm) for _ in range(8)]:
+                f.result()
 
         # per-layer geometry
         self.small, self.tall, self.m_is_tall = [], [], []

@@ -3,6 +3,7 @@
 # torchrun, 1 rank per GPU over RCCL; CPU/gloo when no GPU is visible).
 # ResNet-18 / CIFAR-10 shape, bs 128, lr 0.01, SVD rank 3.
 set -e
+export MIOPEN_FIND_MODE=${MIOPEN_FIND_MODE:-1}
 NPROC=${NPROC:-$(python -c 'import torch;print(max(1,torch.cuda.device_count()))')}
 exec python -m torch.distributed.run --nnodes=1 --nproc-per-node "$NPROC" \
   --master-addr 127.0.0.1 --master-port "${MASTER_PORT:-29517}" \

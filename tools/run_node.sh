@@ -2,6 +2,7 @@
 # Launch PS training on every visible GPU of this node (1 rank per GPU over
 # RCCL).  Usage: tools/run_node.sh [distributed_nn.py args...]
 set -e
+export MIOPEN_FIND_MODE=${MIOPEN_FIND_MODE:-1}
 cd "$(dirname "$0")/.."
 NPROC=${NPROC:-$(python -c 'import torch;print(max(1,torch.cuda.device_count()))')}
 exec python -m torch.distributed.run --nnodes="${NNODES:-1}" \
