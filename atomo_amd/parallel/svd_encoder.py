@@ -145,6 +145,9 @@ class BatchedSVDEncoder:
                 solver_ms = _interp_cost(_SOLVER_EIGH_MS, sm)
                 # pooled host solves overlap ~2 ms of device work for free
                 host_eff = max(0.0, host_ms - 2.0)
+                solver_min = int(os.environ.get("ATOMO_SOLVER_MIN_SM", "0"))
+                if solver_min and sm < solver_min:
+                    continue
                 if solver_ms < host_eff or sm >= 768:
                     solver_dims.add(sm)
             for i, s in enumerate(specs):
