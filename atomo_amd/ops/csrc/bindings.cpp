@@ -6,6 +6,8 @@
 
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime_api.h>
+#include <rocblas/rocblas.h>
+#include <rocsolver/rocsolver.h>
 
 #include <cstdint>
 
@@ -128,6 +130,43 @@ void fused_sgd(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
                          (float)lr, (float)momentum, (float)weight_decay,
                          nesterov, (float)dampening, (float)grad_scale,
                          cur_stream());
+}
+
+// Batched Jacobi eigensolver via rocSOLVER (syevj_strided_batched): solves
+// B symmetric fp32 matrices in one call.  A is overwritten with the
+// eigenvectors in rocBLAS column-major (transpose before use); W ascending.
+// Measured on MI355X: 5-10x SLOWER than hipSOLVER syevd through
+// torch.linalg.eigh (e.g. 34 vs 6.8 ms for 36x256^2), so the encoder keeps
+// syevd; this binding stays for benchmarking alternatives.
+void rocsolver_eigh_batched(torch::Tensor a, torch::Tensor w,
+                            int64_t max_sweeps, double abstol) {
+  check_f32_cuda(a, "a");
+  check_f32_cuda(w, "w");
+  TORCH_CHECK(a.dim() == 3 && a.size(1) == a.size(2), "a must be (B, n, n)");
+  const int B = (int)a.size(0);
+  const int n = (int)a.size(1);
+  TORCH_CHECK(w.numel() == (int64_t)B * n, "w must be (B, n)");
+  static rocblas_handle handle = nullptr;
+  if (handle == nullptr) {
+    TORCH_CHECK(rocblas_create_handle(&handle) == rocblas_status_success,
+                "rocblas_create_handle failed");
+  }
+  rocblas_set_stream(handle, cur_stream());
+  auto opts = torch::TensorOptions()
+                  .device(a.device())
+                  .dtype(torch::kInt32);
+  auto info = torch::zeros({B}, opts);
+  auto n_sweeps = torch::zeros({B}, opts);
+  auto residual = torch::zeros({B}, a.options());
+  auto st = rocsolver_ssyevj_strided_batched(
+      handle, rocblas_esort_ascending, rocblas_evect_original,
+      rocblas_fill_upper, n, a.data_ptr<float>(), n,
+      (rocblas_stride)((int64_t)n * n), (float)abstol,
+      residual.data_ptr<float>(), (rocblas_int)max_sweeps,
+      (rocblas_int*)n_sweeps.data_ptr<int32_t>(), w.data_ptr<float>(),
+      (rocblas_stride)n, (rocblas_int*)info.data_ptr<int32_t>(), B);
+  TORCH_CHECK(st == rocblas_status_success, "rocsolver syevj failed: ",
+              (int)st);
 }
 
 void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor exp_avg,
@@ -326,4 +365,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused rank-k SVD decode+accumulate over W packets");
   m.def("fused_sgd", &fused_sgd, "fused flat SGD apply");
   m.def("fused_adam", &fused_adam, "fused flat Adam apply");
+  m.def("rocsolver_eigh_batched", &rocsolver_eigh_batched,
+        "rocSOLVER batched Jacobi eigensolver (fp32, in-place)");
 }

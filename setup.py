@@ -28,6 +28,7 @@ ext = CUDAExtension(
         "cxx": ["-O3", "-std=c++17"],
         "nvcc": ["-O3", "-std=c++17"],
     },
+    libraries=["rocsolver", "rocblas"],
 )
 
 setup(
