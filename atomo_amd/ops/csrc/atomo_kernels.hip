@@ -4,15 +4,18 @@
 // (reference routines in /root/reference/src/codings/qsgd.py,
 // codings/svd.py, optim/sgd.py — semantics only, no code carried over):
 //
-//   qsgd_pack_kernel       one wave64 per bucket: shfl L2-norm (or terngrad
-//                          clip+max) reduction, counter-hash stochastic
-//                          rounding, LDS-staged (1+q)-bit pack into u32 words.
-//   qsgd_unpack_acc_kernel one thread per packed word, accumulate into the
-//                          PS aggregation buffer.
-//   svd_decode_acc_kernel  fused u·diag(s)·vT over W workers' packets,
-//                          ONE read-modify-write sweep of the output.
-//   fused_sgd_kernel       scale + weight-decay + momentum + apply, one
-//                          flat sweep of the parameter buffer.
+//   qsgd_pack_kernel         one wave64 per bucket: shfl L2-norm (or
+//                            terngrad clip+max) reduction, counter-hash
+//                            stochastic rounding, LDS-staged (1+q)-bit pack.
+//   qsgd_pack/unpack_batched descriptor-table variants: ONE launch covers
+//                            every layer of the model.
+//   qsgd_unpack_acc_kernel   one thread per packed word, accumulate into
+//                            the PS aggregation buffer.
+//   svd_decode_acc_kernel    fused u.diag(s).vT over W workers' packets,
+//                            one read-modify-write sweep of the output.
+//   svd_decode_batched_kernel one launch over all layers x workers.
+//   fused_sgd/adam_kernel    scale + weight-decay + momentum/Adam + apply,
+//                            one flat sweep of the parameter buffer.
 //
 // All kernels are memory-bound sweeps: tiled for 64-wide wavefronts,
 // grid-stride with a grid cap so the 256-CU / 8-XCD chip fills without

@@ -5,11 +5,14 @@
 // ships Grams to the host (reference equivalent: numpy LA.svd per layer,
 // codings/svd.py:95).  Two variants:
 //
-//   jacobi_eigh_kernel      sm <= 64: G and V in LDS (row stride 65 keeps
-//                           column walks conflict-free), 256 threads/WG.
-//   jacobi_eigh_big_kernel  64 < sm <= 512: G stays in its global Gram slot
-//                           (L2-resident: 1 MB for sm=512 fits one XCD's
-//                           4 MB L2), V in a global scratch; 512 threads/WG.
+//   jacobi_eigh_kernel      sm <= 64 (or <= 128 via the JMAX=128 variant):
+//                           G and V in LDS (row stride JMAX+1 keeps column
+//                           walks conflict-free), 256 threads/WG, warm-start
+//                           pre-rotation from the previous step's basis.
+//   jacobi_eigh_big_kernel  global-memory variant (G in its Gram slot, V in
+//                           scratch).  Measured latency-bound and unused by
+//                           default — bigger folds go to batched hipSOLVER
+//                           syevd (see svd_encoder.py routing).
 //
 // Both: each Jacobi round applies all N/2 disjoint plane rotations in two
 // barrier-separated phases (rows = J^T G, then cols = .J and V.J); pairs

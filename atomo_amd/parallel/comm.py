@@ -52,14 +52,6 @@ class Comm:
                     timeout=datetime.timedelta(seconds=300),
                 )
             self._initialized = True
-        self._supports_gather = self.backend != "nccl" or self._nccl_gather_ok()
-
-    @staticmethod
-    def _nccl_gather_ok() -> bool:
-        # ProcessGroupNCCL implements gather via grouped send/recv in
-        # torch >= 1.13; probe the attribute to stay robust.
-        return hasattr(dist, "gather")
-
     # -- collectives -----------------------------------------------------
     def broadcast(self, t: torch.Tensor, src: int = 0) -> None:
         if self._initialized:
@@ -80,22 +72,15 @@ class Comm:
             if out_stacked is not None:
                 out_stacked[0].copy_(send)
             return
-        if self._supports_gather:
-            glist = (
-                [out_stacked[i] for i in range(self.world)]
-                if self.rank == dst
-                else None
-            )
-            dist.gather(send, glist, dst=dst)
-        else:  # pragma: no cover — every supported backend has gather
-            tmp = (
-                out_stacked
-                if self.rank == dst
-                else torch.empty(
-                    self.world, send.numel(), dtype=send.dtype, device=send.device
-                )
-            )
-            dist.all_gather_into_tensor(tmp.view(-1), send)
+        # both gloo (CPU tests) and NCCL/RCCL implement gather; RCCL runs it
+        # as grouped point-to-point sends into rank 0 — exactly the PS
+        # topology's 7-inbound-xGMI-links pattern
+        glist = (
+            [out_stacked[i] for i in range(self.world)]
+            if self.rank == dst
+            else None
+        )
+        dist.gather(send, glist, dst=dst)
 
     def barrier(self) -> None:
         if self._initialized:
