@@ -49,6 +49,10 @@ SOLVER_SM = 4096
 # measured per-matrix costs (ms) on MI355X + EPYC host, fp32
 _HOST_EIGH_MS = {128: 0.9, 256: 3.3, 512: 10.7, 1024: 45.0, 2048: 160.0}
 _SOLVER_EIGH_MS = {128: 2.1, 256: 5.6, 512: 11.5, 1024: 21.0, 2048: 44.0}
+# hipSOLVER syevd cost model: per-call base + per-extra-matrix increment
+# (measured: 512 B=12 is 12.6 ms vs 10.1 at B=1; 1024 B=7 is 25.8 vs 19.4)
+_SYEVD_BASE = {256: 5.2, 512: 10.1, 1024: 19.4, 2048: 41.8}
+_SYEVD_INCR = {256: 0.05, 512: 0.25, 1024: 1.1, 2048: 6.5}
 
 
 def _interp_cost(table, sm):
@@ -369,40 +373,53 @@ class BatchedSVDEncoder:
                 # (serially they cost e.g. 5.6+11.5+21 ms on ResNet-50);
                 # results written back into the gram slots / evals buffer
                 if self.solver_layers:
+                    # group by fold size, then MERGE small groups into the
+                    # largest group's call by zero-padding when the syevd
+                    # cost model says the per-matrix increment at the big
+                    # size is cheaper than another call's base.  Padding a
+                    # PSD Gram with zero rows/cols only appends zero
+                    # eigenvalues, so the top-sm (descending) eigenpairs are
+                    # the original spectrum and their vectors have zero pad
+                    # components.
                     by_sm = defaultdict(list)
                     for i in self.solver_layers:
                         by_sm[self.small[i]].append(i)
-                    if not hasattr(self, "_solver_streams"):
-                        self._solver_streams = [
-                            torch.cuda.Stream() for _ in range(len(by_sm))
-                        ]
-                    main = torch.cuda.current_stream()
-                    gram_ready = torch.cuda.Event()
-                    gram_ready.record(main)
-                    for st, (sm, idxs) in zip(self._solver_streams, by_sm.items()):
-                        with torch.cuda.stream(st):
-                            st.wait_event(gram_ready)
-                            gs = torch.stack(
-                                [
-                                    self.grams[
-                                        self.gram_offsets[i] : self.gram_offsets[i]
-                                        + sm * sm
-                                    ].view(sm, sm)
-                                    for i in idxs
-                                ]
-                            )
-                            gs = 0.5 * (gs + gs.transpose(1, 2))
-                            evals, evecs = torch.linalg.eigh(gs)
-                            evals = evals.flip(1).clamp(min=0.0)
-                            evecs = evecs.flip(2)
-                            for j, i in enumerate(idxs):
-                                o = self.eval_offs[self.layer_row[i]]
-                                self.evals_dev[o : o + sm].copy_(evals[j])
-                                self.grams[
-                                    self.gram_offsets[i] : self.gram_offsets[i]
-                                    + sm * sm
-                                ].copy_(evecs[j].reshape(-1))
-                        main.wait_stream(st)
+                    sizes = sorted(by_sm, reverse=True)
+                    n_top = sizes[0]
+                    calls = {n_top: list(by_sm[n_top])}
+                    for sm in sizes[1:]:
+                        cnt = len(by_sm[sm])
+                        sep = _interp_cost(_SYEVD_BASE, sm) + _interp_cost(
+                            _SYEVD_INCR, sm
+                        ) * (cnt - 1)
+                        merged = _interp_cost(_SYEVD_INCR, n_top) * cnt
+                        if merged < sep:
+                            calls[n_top].extend(by_sm[sm])
+                        else:
+                            calls[sm] = list(by_sm[sm])
+                    for n_call, idxs in calls.items():
+                        B = len(idxs)
+                        gs = torch.zeros(
+                            B, n_call, n_call, device=self.device
+                        )
+                        for j, i in enumerate(idxs):
+                            sm = self.small[i]
+                            gs[j, :sm, :sm] = self.grams[
+                                self.gram_offsets[i] : self.gram_offsets[i]
+                                + sm * sm
+                            ].view(sm, sm)
+                        gs = 0.5 * (gs + gs.transpose(1, 2))
+                        evals, evecs = torch.linalg.eigh(gs)
+                        evals = evals.flip(1).clamp(min=0.0)
+                        evecs = evecs.flip(2)
+                        for j, i in enumerate(idxs):
+                            sm = self.small[i]
+                            o = self.eval_offs[self.layer_row[i]]
+                            self.evals_dev[o : o + sm].copy_(evals[j, :sm])
+                            self.grams[
+                                self.gram_offsets[i] : self.gram_offsets[i]
+                                + sm * sm
+                            ].copy_(evecs[j, :sm, :sm].reshape(-1))
                 self.evals_host.copy_(self.evals_dev, non_blocking=True)
             elif host_layers:
                 grams_host = self.grams.to("cpu")  # synchronous copy
