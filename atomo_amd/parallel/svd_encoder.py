@@ -227,6 +227,101 @@ class BatchedSVDEncoder:
             else:
                 self.use_kernels = False
 
+    # -- solver-eigh overlap with backward -----------------------------
+    # Deep layers' gradients materialize FIRST in backward (autograd walks
+    # output->input), and the expensive hipSOLVER eighs belong to exactly
+    # those 1x1-conv folds.  In overlap mode the per-layer gram hooks count
+    # down each solver call-group; when a group's last Gram is on the side
+    # stream, a pool thread launches its (host-blocking) batched eigh on a
+    # dedicated stream — hiding most of its cost under the rest of backward.
+    def build_solver_plan(self):
+        """(call_size -> layer list) using the same merge rule as encode."""
+        if not self.solver_layers:
+            return {}
+        by_sm = defaultdict(list)
+        for i in self.solver_layers:
+            by_sm[self.small[i]].append(i)
+        sizes = sorted(by_sm, reverse=True)
+        n_top = sizes[0]
+        calls = {n_top: list(by_sm[n_top])}
+        for sm in sizes[1:]:
+            cnt = len(by_sm[sm])
+            sep = _interp_cost(_SYEVD_BASE, sm) + _interp_cost(_SYEVD_INCR, sm) * (
+                cnt - 1
+            )
+            merged = _interp_cost(_SYEVD_INCR, n_top) * cnt
+            if merged < sep:
+                calls[n_top].extend(by_sm[sm])
+            else:
+                calls[sm] = list(by_sm[sm])
+        return calls
+
+    def setup_solver_overlap(self, side_stream) -> None:
+        self._ov_side = side_stream
+        self._ov_stream = torch.cuda.Stream()
+        self._ov_plan = self.build_solver_plan()
+        self._ov_group_of = {}
+        for n_call, idxs in self._ov_plan.items():
+            for i in idxs:
+                self._ov_group_of[i] = n_call
+        self._ov_pending = {}
+        self._ov_futures = []
+
+    def arm_overlap(self) -> None:
+        if getattr(self, "_ov_plan", None):
+            self._ov_pending = {n: set(idxs) for n, idxs in self._ov_plan.items()}
+            self._ov_futures = []
+
+    def _solver_call(self, n_call, idxs, ready_evt) -> None:
+        with torch.cuda.stream(self._ov_stream):
+            self._ov_stream.wait_event(ready_evt)
+            self._run_solver_group(n_call, idxs)
+
+    def _run_solver_group(self, n_call, idxs) -> None:
+        B = len(idxs)
+        gs = torch.zeros(B, n_call, n_call, device=self.device)
+        for j, i in enumerate(idxs):
+            sm = self.small[i]
+            gs[j, :sm, :sm] = self.grams[
+                self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
+            ].view(sm, sm)
+        gs = 0.5 * (gs + gs.transpose(1, 2))
+        evals, evecs = torch.linalg.eigh(gs)
+        evals = evals.flip(1).clamp(min=0.0)
+        evecs = evecs.flip(2)
+        for j, i in enumerate(idxs):
+            sm = self.small[i]
+            o = self.eval_offs[self.layer_row[i]]
+            self.evals_dev[o : o + sm].copy_(evals[j, :sm])
+            self.grams[
+                self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
+            ].copy_(evecs[j, :sm, :sm].reshape(-1))
+
+    def on_overlap_gram_done(self, i) -> None:
+        """Called from the gram hook (inside the side-stream context)."""
+        n_call = self._ov_group_of.get(i)
+        if n_call is None or not self._ov_pending:
+            return
+        pend = self._ov_pending.get(n_call)
+        if pend is None:
+            return
+        pend.discard(i)
+        if not pend:
+            del self._ov_pending[n_call]
+            evt = torch.cuda.Event()
+            evt.record(self._ov_side)
+            idxs = self._ov_plan[n_call]
+            self._ov_futures.append(
+                self._pool.submit(self._solver_call, n_call, idxs, evt)
+            )
+
+    def finish_overlap_solvers(self) -> None:
+        for f in getattr(self, "_ov_futures", []):
+            f.result()  # hipSOLVER blocks its thread: done == complete
+        self._ov_futures = []
+        if getattr(self, "_ov_plan", None):
+            torch.cuda.current_stream().wait_stream(self._ov_stream)
+
     # -----------------------------------------------------------------
     def _a2d(self, grad: torch.Tensor, spec: LayerSpec) -> torch.Tensor:
         m, n = spec.meta["m"], spec.meta["n"]
@@ -313,6 +408,8 @@ class BatchedSVDEncoder:
                 marks.append(time.perf_counter())
                 print(f"[enc] {label}: {1e3*(marks[-1]-marks[-2]):.2f} ms", flush=True)
 
+        if grams_done and getattr(self, "_ov_plan", None):
+            self.finish_overlap_solvers()
         host_layers = [i for i in range(len(specs)) if i not in kernel_set]
         mm_layers = host_layers + (self.big_gram_layers if use_kernels else [])
         a2ds = {i: self._a2d(grads[i], specs[i]) for i in mm_layers}
@@ -372,7 +469,9 @@ class BatchedSVDEncoder:
                 # group on its own stream so independent solves overlap
                 # (serially they cost e.g. 5.6+11.5+21 ms on ResNet-50);
                 # results written back into the gram slots / evals buffer
-                if self.solver_layers:
+                if self.solver_layers and not (
+                    grams_done and getattr(self, "_ov_plan", None)
+                ):
                     # group by fold size, then MERGE small groups into the
                     # largest group's call by zero-padding when the syevd
                     # cost model says the per-matrix increment at the big

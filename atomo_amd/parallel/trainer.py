@@ -182,6 +182,8 @@ class PSTrainer:
         if self.is_worker:
             with t.phase("comp"):
                 self.model.train()
+                if self.overlap:
+                    self.wc.arm_overlap()
                 if self.use_graph:
                     self._fwd_bwd_graphed(x, y)
                 else:

@@ -144,6 +144,7 @@ class WireCodec:
             return False
         if isinstance(self.codec, SVDCodec) and self._batched_encoder is not None:
             enc = self._batched_encoder
+            enc.setup_solver_overlap(side_stream)
 
             def make_hook(i):
                 spec = self.specs[i]
@@ -163,6 +164,7 @@ class WireCodec:
                             _t.mm(a.t(), a, out=gv)
                         else:
                             _t.mm(a, a.t(), out=gv)
+                        enc.on_overlap_gram_done(i)
 
                 return hook
 
@@ -197,6 +199,10 @@ class WireCodec:
                 p.register_post_accumulate_grad_hook(make_hook(i))
             return True
         return False
+
+    def arm_overlap(self) -> None:
+        if self._batched_encoder is not None:
+            self._batched_encoder.arm_overlap()
 
     # -- worker side -----------------------------------------------------
     def encode_all(
