@@ -264,13 +264,22 @@ class BatchedSVDEncoder:
         for n_call, idxs in self._ov_plan.items():
             for i in idxs:
                 self._ov_group_of[i] = n_call
+        # host-LAPACK layers solve from hooks too: their Gram D2H + eigh
+        # rides a pool thread while backward continues
+        self._ov_host_set = {
+            i for i in range(len(self.specs)) if i not in self.kernel_set
+        }
         self._ov_pending = {}
         self._ov_futures = []
+        self._ov_host_results = {}
+        self._ov_armed = False
 
     def arm_overlap(self) -> None:
-        if getattr(self, "_ov_plan", None):
+        if getattr(self, "_ov_plan", None) is not None:
             self._ov_pending = {n: set(idxs) for n, idxs in self._ov_plan.items()}
             self._ov_futures = []
+            self._ov_host_results = {}
+            self._ov_armed = True
 
     def _solver_call(self, n_call, idxs, ready_evt) -> None:
         with torch.cuda.stream(self._ov_stream):
@@ -297,8 +306,34 @@ class BatchedSVDEncoder:
                 self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
             ].copy_(evecs[j, :sm, :sm].reshape(-1))
 
+    def _host_solve_layer(self, i, ready_evt) -> None:
+        ready_evt.synchronize()
+        sm = self.small[i]
+        g = (
+            self.grams[self.gram_offsets[i] : self.gram_offsets[i] + sm * sm]
+            .view(sm, sm)
+            .to("cpu")
+        )
+        dt = torch.float32 if sm >= 96 else torch.float64
+        g = g.to(dt)
+        g = 0.5 * (g + g.t())
+        evals, evecs = torch.linalg.eigh(g)
+        self._ov_host_results[i] = (
+            evals.flip(0).clamp(min=0.0).to(torch.float64).sqrt(),
+            evecs.flip(1).to(torch.float64),
+        )
+
     def on_overlap_gram_done(self, i) -> None:
         """Called from the gram hook (inside the side-stream context)."""
+        if not getattr(self, "_ov_armed", False):
+            return
+        if i in self._ov_host_set:
+            evt = torch.cuda.Event()
+            evt.record(self._ov_side)
+            self._ov_futures.append(
+                self._pool.submit(self._host_solve_layer, i, evt)
+            )
+            return
         n_call = self._ov_group_of.get(i)
         if n_call is None or not self._ov_pending:
             return
@@ -408,9 +443,13 @@ class BatchedSVDEncoder:
                 marks.append(time.perf_counter())
                 print(f"[enc] {label}: {1e3*(marks[-1]-marks[-2]):.2f} ms", flush=True)
 
-        if grams_done and getattr(self, "_ov_plan", None):
+        ov_host = {}
+        if grams_done and getattr(self, "_ov_plan", None) is not None:
             self.finish_overlap_solvers()
+            self._ov_armed = False
+            ov_host = self._ov_host_results
         host_layers = [i for i in range(len(specs)) if i not in kernel_set]
+        host_unsolved = [i for i in host_layers if i not in ov_host]
         mm_layers = host_layers + (self.big_gram_layers if use_kernels else [])
         a2ds = {i: self._a2d(grads[i], specs[i]) for i in mm_layers}
 
@@ -447,7 +486,7 @@ class BatchedSVDEncoder:
         gram_event = None
         if self.device.type == "cuda":
             if use_kernels:
-                if host_layers:
+                if host_unsolved:
                     self.grams_host.copy_(self.grams, non_blocking=True)
                     grams_host = self.grams_host
                     gram_event = torch.cuda.Event()
@@ -566,10 +605,13 @@ class BatchedSVDEncoder:
             evecs = evecs.flip(2).to(torch.float64)
             return idxs, evals, evecs
 
+        for i, (sv, ev) in ov_host.items():
+            svals_h[i] = sv
+            evecs_h[i] = ev
         futures = []
-        if host_layers:
+        if host_unsolved:
             by_dim = defaultdict(list)
-            for i in host_layers:
+            for i in host_unsolved:
                 by_dim[self.small[i]].append(i)
             # chunk big-sm groups so the pool actually parallelizes them
             items = []
