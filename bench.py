@@ -44,8 +44,10 @@ def main():
                    help="capture forward/backward in a hipGraph and replay")
     p.add_argument("--channels-last", action="store_true", default=False,
                    help="NHWC memory format for convs")
-    p.add_argument("--overlap", action="store_true", default=False,
-                   help="backward-hook per-layer encode on a side stream")
+    p.add_argument("--overlap", action="store_true", default=None,
+                   help="backward-hook per-layer encode on a side stream "
+                        "(default: on for svd, off otherwise)")
+    p.add_argument("--no-overlap", dest="overlap", action="store_false")
     p.add_argument("--amp", action="store_true", default=False,
                    help="bf16 autocast fwd/bwd (reported dtype changes)")
     a = p.parse_args()
@@ -80,7 +82,7 @@ def main():
         seed=42,
         device=device,
         use_graph=a.graph,
-        overlap=a.overlap,
+        overlap=(a.code == "svd") if a.overlap is None else a.overlap,
         defer_loss=True,
         amp=a.amp,
     )
