@@ -307,13 +307,18 @@ class BatchedSVDEncoder:
             ].copy_(evecs[j, :sm, :sm].reshape(-1))
 
     def _host_solve_layer(self, i, ready_evt) -> None:
-        ready_evt.synchronize()
+        ready_evt.synchronize()  # gram data final (host-side wait)
         sm = self.small[i]
-        g = (
-            self.grams[self.gram_offsets[i] : self.gram_offsets[i] + sm * sm]
-            .view(sm, sm)
-            .to("cpu")
-        )
+        with torch.cuda.stream(self._ov_stream):
+            # D2H on the overlap stream: the default stream is busy with
+            # backward and would delay this copy by the whole queue
+            g = (
+                self.grams[
+                    self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
+                ]
+                .view(sm, sm)
+                .to("cpu")
+            )
         dt = torch.float32 if sm >= 96 else torch.float64
         g = g.to(dt)
         g = 0.5 * (g + g.t())
