@@ -46,10 +46,13 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
     p.add_argument("--data-root", type=str, default=None,
                    help="directory with the real dataset files (standard "
                         "binary formats); synthetic data when absent")
-    p.add_argument("--comm-type", type=str, default="Bcast")
+    p.add_argument("--comm-type", type=str, default="Bcast",
+                   help="Bcast = collective gather; P2P = arrival-order "
+                        "isend/irecv with decode-as-arrives")
     p.add_argument("--num-aggregate", type=int, default=0,
-                   help="gradients to collect per step (0 = all workers; the "
-                        "reference stores but never uses this flag)")
+                   help="with --comm-type P2P: aggregate only the first K "
+                        "arriving gradients (0 = all; the reference stores "
+                        "this flag but never implements it)")
     p.add_argument("--eval-freq", type=int, default=0)
     p.add_argument("--train-dir", type=str, default="output/models/")
     p.add_argument("--compress", type=_ref_bool, nargs="?", const=True, default=True)
@@ -124,6 +127,8 @@ class RunConfig:
             train_dir=a.train_dir,
             use_graph=a.graph,
             overlap=a.overlap,
+            comm_type=a.comm_type,
+            num_aggregate=a.num_aggregate,
             step_timeout=a.step_timeout,
             amp=a.amp,
         )

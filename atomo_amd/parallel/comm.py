@@ -82,6 +82,53 @@ class Comm:
         )
         dist.gather(send, glist, dst=dst)
 
+    def gather_arrival(
+        self,
+        send: torch.Tensor,
+        out_stacked: Optional[torch.Tensor],
+        dst: int = 0,
+        on_arrival=None,
+    ):
+        """Point-to-point gather with ARRIVAL-ORDER notification — the
+        reference's `MPI.Request.waitany` drain loop
+        (sync_replicas_master_nn.py:198-215) mapped onto isend/irecv:
+        the PS calls ``on_arrival(rank)`` as each worker's bucket lands
+        (its own row first), so decode overlaps the remaining receives.
+        Returns the arrival order on dst, else None."""
+        if not self._initialized:
+            if out_stacked is not None:
+                out_stacked[0].copy_(send)
+                if on_arrival is not None:
+                    on_arrival(0)
+            return [0] if out_stacked is not None else None
+        if self.rank == dst:
+            out_stacked[dst].copy_(send)
+            reqs = {
+                w: dist.irecv(out_stacked[w], src=w)
+                for w in range(self.world)
+                if w != dst
+            }
+            order = [dst]
+            if on_arrival is not None:
+                on_arrival(dst)
+            pending = dict(reqs)
+            while pending:
+                done = [w for w, r in pending.items() if r.is_completed()]
+                if not done:
+                    # block on the longest-outstanding request
+                    w = next(iter(pending))
+                    pending.pop(w).wait()
+                    done = [w]
+                    done += [v for v, r in pending.items() if r.is_completed()]
+                for w in done:
+                    pending.pop(w, None)
+                    order.append(w)
+                    if on_arrival is not None:
+                        on_arrival(w)
+            return order
+        dist.isend(send, dst=dst).wait()
+        return None
+
     def barrier(self) -> None:
         if self._initialized:
             if self.backend == "nccl":

@@ -63,6 +63,8 @@ class PSTrainer:
         step_timeout: float = 0.0,
         defer_loss: bool = False,
         amp: bool = False,
+        comm_type: str = "Bcast",
+        num_aggregate: int = 0,
     ):
         self.comm = comm
         self.device = device or comm.device
@@ -80,6 +82,14 @@ class PSTrainer:
         self.loss_fn = nn.CrossEntropyLoss()
 
         self.dedicated_ps = dedicated_ps and comm.world > 1
+        # P2P comm mode: arrival-order gather + decode-as-arrives (the
+        # reference's waitany drain, sync_replicas_master_nn.py:198-215);
+        # num_aggregate > 0 aggregates only the first K arrivals (the
+        # reference stores this flag but never implements it — here the
+        # PS drops late gradients from the update, full-sync timing).
+        self.p2p = comm_type.lower() in ("p2p", "isend") and comm.world > 1
+        self.num_aggregate = int(num_aggregate or 0)
+        self._last_contrib = 0
         self.is_master = comm.rank == 0
         self.is_worker = (not self.dedicated_ps) or comm.rank > 0
         self.num_workers = comm.world - 1 if self.dedicated_ps else comm.world
@@ -111,6 +121,8 @@ class PSTrainer:
             if self.is_master and not self.wc.reducible and comm.world > 1
             else None
         )
+        if self.p2p and self.wc.reducible:
+            self.p2p = False  # raw codec always rides the RCCL reduce
         self.agg = (
             torch.zeros_like(self.flat)
             if self.is_master and not self.wc.reducible
@@ -216,25 +228,54 @@ class PSTrainer:
         elif self.wc.reducible:
             self.flat_grad.zero_()  # dedicated PS contributes zeros to the sum
 
-        with t.phase("comm"):
-            if self.wc.reducible:
-                self.comm.reduce_sum(self.wire, dst=0)
-            elif self.comm.world > 1:
-                self.comm.gather(self.wire, self.gather_buf, dst=0)
+        if self.p2p and not self.wc.reducible:
+            # arrival-order gather: the PS decodes each worker's bucket as
+            # it lands, overlapping decode with the remaining receives
+            with t.phase("comm"):
+                contrib = [0]
+                target = self.num_aggregate or self.num_workers
+                if self.is_master:
+                    self.agg.zero_()
+
+                    def on_arrival(w):
+                        if self.dedicated_ps and w == 0:
+                            return
+                        if contrib[0] < target:
+                            self.wc.decode_all(
+                                self.gather_buf[w : w + 1], self.agg
+                            )
+                            contrib[0] += 1
+
+                    self.comm.gather_arrival(
+                        self.wire, self.gather_buf, dst=0, on_arrival=on_arrival
+                    )
+                else:
+                    self.comm.gather_arrival(self.wire, None, dst=0)
+            self._last_contrib = contrib[0]
+            grad_flat = self.agg if self.is_master else None
+        else:
+            with t.phase("comm"):
+                if self.wc.reducible:
+                    self.comm.reduce_sum(self.wire, dst=0)
+                elif self.comm.world > 1:
+                    self.comm.gather(self.wire, self.gather_buf, dst=0)
+
+            if self.is_master:
+                with t.phase("decode"):
+                    if self.wc.reducible:
+                        grad_flat = self.wire
+                    else:
+                        self.agg.zero_()
+                        rows = (
+                            self.gather_buf
+                            if self.comm.world > 1
+                            else self.wire.view(1, -1)
+                        )
+                        self.wc.decode_all(rows, self.agg)
+                        grad_flat = self.agg
+            self._last_contrib = self.num_workers
 
         if self.is_master:
-            with t.phase("decode"):
-                if self.wc.reducible:
-                    grad_flat = self.wire
-                else:
-                    self.agg.zero_()
-                    rows = (
-                        self.gather_buf
-                        if self.comm.world > 1
-                        else self.wire.view(1, -1)
-                    )
-                    self.wc.decode_all(rows, self.agg)
-                    grad_flat = self.agg
             with t.phase("apply"):
                 self.opt.lr = self.lr
                 self._apply(grad_flat)
@@ -296,7 +337,7 @@ class PSTrainer:
         self.last_loss = float(self._static_loss)
 
     def _apply(self, grad_flat: torch.Tensor) -> None:
-        scale = 1.0 / max(1, self.num_workers)
+        scale = 1.0 / max(1, self._last_contrib or self.num_workers)
         if self.flat.is_cuda and type(self.opt).__name__ == "ExternalSGD":
             from ..ops import optim_ops
 

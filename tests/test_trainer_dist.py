@@ -88,3 +88,67 @@ def test_dist_dedicated_ps():
     assert all(math.isnan(l) for l in results[0][0])
     assert not any(math.isnan(l) for l in results[1][0])
     assert results[0][1] == pytest.approx(results[1][1], rel=1e-5)
+
+
+def _run_rank_p2p(rank, world, port, num_agg, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.set_num_threads(2)
+    torch.manual_seed(100 + rank)
+    from atomo_amd.codings import make_codec
+    from atomo_amd.data import make_loaders
+    from atomo_amd.parallel import Comm, PSTrainer
+
+    comm = Comm(backend="gloo", device=torch.device("cpu"))
+    trainer = PSTrainer(
+        model_name="LeNet",
+        codec=make_codec("svd", rank=3),
+        comm=comm, lr=0.05, momentum=0.9, num_classes=10, in_channels=1,
+        seed=7, device=torch.device("cpu"), comm_type="P2P",
+        num_aggregate=num_agg,
+    )
+    assert trainer.p2p
+    train, _ = make_loaders("mnist", 16, 16, torch.device("cpu"), seed=50 + rank)
+    losses = []
+    for i, (x, y) in enumerate(train):
+        losses.append(trainer.train_step(x, y))
+        if i >= 7:
+            break
+    trainer.comm.broadcast(trainer.flat, src=0)
+    q.put((rank, losses, trainer.flat.sum().item(),
+           trainer._last_contrib if rank == 0 else -1))
+    comm.barrier()
+    comm.close()
+
+
+@pytest.mark.parametrize("num_agg,port", [(0, 29621), (2, 29622)])
+def test_dist_p2p_mode(num_agg, port):
+    """Arrival-order P2P gather: full and partial (num_aggregate)
+    aggregation both train and keep ranks weight-synchronized."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    world = 3
+    procs = [
+        ctx.Process(target=_run_rank_p2p, args=(r, world, port, num_agg, q))
+        for r in range(world)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, losses, fsum, contrib = q.get(timeout=180)
+        results[rank] = (losses, fsum, contrib)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    import math
+
+    assert not any(math.isnan(l) for l in results[0][0])
+    # ranks converge to the same weights
+    assert results[0][1] == pytest.approx(results[1][1], rel=1e-5)
+    assert results[0][1] == pytest.approx(results[2][1], rel=1e-5)
+    # contribution count: all 3 (colocated) or capped at num_aggregate
+    expect = 3 if num_agg == 0 else num_agg
+    assert results[0][2] == expect
