@@ -55,6 +55,43 @@ _SYEVD_BASE = {256: 5.2, 512: 10.1, 1024: 19.4, 2048: 41.8}
 _SYEVD_INCR = {256: 0.05, 512: 0.25, 1024: 1.1, 2048: 6.5}
 
 
+def _robust_eigh(g: torch.Tensor, out_dtype=torch.float64):
+    """torch.linalg.eigh with convergence fallbacks.
+
+    Gradient Grams are often massively rank-deficient (grad rank <= batch
+    size), and MKL/hipSOLVER syevd intermittently fails to converge on
+    them in fp32.  Ladder: fp32/as-given -> fp64 -> fp64 + diagonal
+    jitter -> zero spectrum (degenerate packets ship atom 0)."""
+    try:
+        evals, evecs = torch.linalg.eigh(g)
+        return evals.to(out_dtype), evecs.to(out_dtype)
+    except Exception:
+        pass
+    g64 = g.to(torch.float64)
+    try:
+        evals, evecs = torch.linalg.eigh(g64)
+        return evals.to(out_dtype), evecs.to(out_dtype)
+    except Exception:
+        pass
+    n = g64.shape[-1]
+    scale = g64.diagonal(dim1=-2, dim2=-1).abs().amax(dim=-1, keepdim=True)
+    jitter = (scale.clamp(min=1e-30) * 1e-6).unsqueeze(-1) * torch.eye(
+        n, dtype=torch.float64, device=g64.device
+    )
+    try:
+        evals, evecs = torch.linalg.eigh(g64 + jitter)
+        return evals.to(out_dtype), evecs.to(out_dtype)
+    except Exception:
+        shape = g64.shape
+        evals = torch.zeros(shape[:-1], dtype=out_dtype, device=g64.device)
+        evecs = (
+            torch.eye(n, dtype=out_dtype, device=g64.device)
+            .expand(shape)
+            .clone()
+        )
+        return evals, evecs
+
+
 def _interp_cost(table, sm):
     ks = sorted(table)
     for k in ks:
@@ -295,7 +332,7 @@ class BatchedSVDEncoder:
                 self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
             ].view(sm, sm)
         gs = 0.5 * (gs + gs.transpose(1, 2))
-        evals, evecs = torch.linalg.eigh(gs)
+        evals, evecs = _robust_eigh(gs, out_dtype=torch.float32)
         evals = evals.flip(1).clamp(min=0.0)
         evecs = evecs.flip(2)
         for j, i in enumerate(idxs):
@@ -322,10 +359,10 @@ class BatchedSVDEncoder:
         dt = torch.float32 if sm >= 96 else torch.float64
         g = g.to(dt)
         g = 0.5 * (g + g.t())
-        evals, evecs = torch.linalg.eigh(g)
+        evals, evecs = _robust_eigh(g)
         self._ov_host_results[i] = (
-            evals.flip(0).clamp(min=0.0).to(torch.float64).sqrt(),
-            evecs.flip(1).to(torch.float64),
+            evals.flip(0).clamp(min=0.0).sqrt(),
+            evecs.flip(1),
         )
 
     def on_overlap_gram_done(self, i) -> None:
@@ -552,7 +589,7 @@ class BatchedSVDEncoder:
                                 + sm * sm
                             ].view(sm, sm)
                         gs = 0.5 * (gs + gs.transpose(1, 2))
-                        evals, evecs = torch.linalg.eigh(gs)
+                        evals, evecs = _robust_eigh(gs, out_dtype=torch.float32)
                         evals = evals.flip(1).clamp(min=0.0)
                         evecs = evecs.flip(2)
                         for j, i in enumerate(idxs):
@@ -605,9 +642,9 @@ class BatchedSVDEncoder:
                 ]
             ).to(dt)
             gs = 0.5 * (gs + gs.transpose(1, 2))
-            evals, evecs = torch.linalg.eigh(gs)
-            evals = evals.flip(1).clamp(min=0.0).to(torch.float64)
-            evecs = evecs.flip(2).to(torch.float64)
+            evals, evecs = _robust_eigh(gs)
+            evals = evals.flip(1).clamp(min=0.0)
+            evecs = evecs.flip(2)
             return idxs, evals, evecs
 
         for i, (sv, ev) in ov_host.items():
