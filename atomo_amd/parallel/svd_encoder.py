@@ -1,19 +1,28 @@
-"""Batched SVD wire encoder — ONE host sync per step for the whole model.
+"""Batched SVD wire encoder — at most one host sync per step, zero in the
+fully-device configuration.
 
 The naive per-layer path costs ~60 host round trips per step (factorize,
-sample, write).  This encoder restructures the work MI355X-first; layers
-split into two classes:
+sample, write).  This encoder restructures the work MI355X-first:
 
-DEVICE layers (even 2-D fold, wire budget r_max <= 32, small-dim <= 512 —
-every layer of the BASELINE model configs):
-  batched_gram (hand-written kernel, small-dim <= 64) or rocBLAS Gram
-  (bigger folds) -> jacobi_eigh / jacobi_eigh_big (parallel cyclic Jacobi:
-  LDS-resident for sm <= 64, L2-resident global for 64 < sm <= 512) ->
-  [evals to host: the ONLY sync] -> vectorized host Bernoulli sampling ->
-  sel_table H2D -> build_stage -> batched_sel writes the wire packets.
-HOST layers (odd zero-padded 1-D folds, or r_max > 32 / sm > 512):
-  rocBLAS Gram, fp64/fp32 LAPACK eigh on a thread pool, staged factors H2D,
-  rocBLAS selection GEMMs straight into the wire.
+DEVICE layers (even 2-D fold, wire budget r_max <= 32):
+  * small-dim <= 64: batched_gram kernel -> warm-started LDS Jacobi
+    (jacobi_eigh_kernel, previous step's eigenbasis pre-rotates so sweeps
+    converge in ~1-2 iterations).
+  * bigger folds (1x1-conv Grams up to 2048x1024): rocBLAS Gram ->
+    batched hipSOLVER syevd, size-groups merged into one call by
+    zero-padding when the measured cost model favors it.
+  * then: fused on-device Bernoulli sampler + stage builder
+    (sample_stage_kernel; counter-hash RNG; Msg bytes counted on device)
+    -> batched_sel kernel writes the wire packets.  No host sync at all
+    when no host layers exist and no host RNG generator is pinned.
+HOST layers (odd zero-padded 1-D folds, r_max > 32, or sm > 4096):
+  rocBLAS Gram, robust LAPACK eigh on a warmed thread pool (overlapping
+  the device work), host sampling, staged factors H2D, rocBLAS selection
+  GEMMs straight into the wire.
+OVERLAP mode (--overlap): per-layer Gram hooks fire during backward; each
+  solver call-group's eigh and each host layer's LAPACK solve launch from a
+  pool thread the moment their last Gram lands — deep layers' gradients
+  materialize first, so most eigensolve time hides under backward.
 
 Semantics identical to SVDCodec.encode_into (same wire layout, same
 sampler, same unbiasedness invariant E[sum s_i/p_i u_i v_i^T] = grad;
