@@ -15,7 +15,7 @@ Wire layout per layer (fp32 words, int words bit-cast):
     [norms (n_buckets)] [packed (n_buckets * words_per_bucket) int32]
 
 Hot path on MI355X: the pack/unpack loops are the HIP kernels in
-ops/csrc/qsgd_kernels.hip (one wave per bucket: DPP/shfl L2-norm reduction,
+ops/csrc/atomo_kernels.hip (one wave per bucket: DPP/shfl L2-norm reduction,
 philox stochastic rounding, shift-or pack — reference hot spots SURVEY §2.10).
 The torch implementation below is the CPU oracle those kernels are tested
 against.

@@ -1,4 +1,4 @@
-"""Fused SGD apply (kernel: ops/csrc/optim_kernels.hip).
+"""Fused SGD apply (kernel: ops/csrc/atomo_kernels.hip).
 
 One sweep over the flat parameter buffer:
     g    = grad * grad_scale + weight_decay * p

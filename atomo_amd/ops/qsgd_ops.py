@@ -1,4 +1,4 @@
-"""QSGD pack/unpack GPU wrappers (kernels: ops/csrc/qsgd_kernels.hip).
+"""QSGD pack/unpack GPU wrappers (kernels: ops/csrc/atomo_kernels.hip).
 
 Wire layout per layer (inside an fp32 region, int words bit-cast):
     [norms (n_buckets fp32)] [packed (n_buckets * words_per_bucket) u32]

@@ -10,7 +10,7 @@ eigensolve (n <= 64 for every conv layer) touches the host.
 
 ``decode_acc``: fused rank-k reconstruction u.diag(s).vT of MANY packets
 accumulated into the PS aggregation buffer in ONE output sweep (kernel:
-ops/csrc/svd_kernels.hip) — W workers' packets cost one read-modify-write
+ops/csrc/atomo_kernels.hip) — W workers' packets cost one read-modify-write
 of the output instead of W.
 """
 

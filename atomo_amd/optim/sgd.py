@@ -3,7 +3,7 @@
 Math identical to reference optim/sgd.py:57-89 (momentum buffer
 ``buf = mu*buf + d_p``, optional nesterov / weight decay) but the gradient is
 one flat fp32 tensor (the PS aggregation buffer) and the whole update is a
-single fused HIP kernel on MI355X (ops/csrc/optim_kernels.hip) — the
+single fused HIP kernel on MI355X (ops/csrc/atomo_kernels.hip) — the
 reference loops per-parameter through numpy (SURVEY §2.10 row "SGD apply").
 """
 
