@@ -13,7 +13,7 @@ MI355X-first differences from the reference:
     so a whole model's packets ship as one RCCL gather (no pickle).
   * decode is a fused rank-k accumulate (``out2d += (u*s) @ vT``) straight
     into the PS aggregation buffer — hipBLASLt GEMM or the HIP kernel in
-    ops/csrc/svd_kernels.hip, never a host hop.
+    ops/csrc/atomo_kernels.hip, never a host hop.
   * factorization backend is pluggable: ``torch`` (torch.linalg.svd, the
     oracle) or ``gram`` (one-pass Gram + Jacobi eigensolver on MFMA — the
     tall-skinny shapes here have n <= ~64, so A^T A is tiny and exact).
