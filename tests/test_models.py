@@ -47,3 +47,12 @@ def test_zoo_covers_reference_families():
     for required in {"LeNet", "FC", "ResNet18", "ResNet34", "ResNet50",
                      "ResNet101", "ResNet152", "VGG11", "AlexNet", "DenseNet"}:
         assert required in names
+
+
+@pytest.mark.parametrize("name", ["ResNet34", "ResNet101", "VGG16", "VGG19"])
+def test_deep_zoo_forward(name):
+    torch.manual_seed(0)
+    model = build_model(name, num_classes=10, in_channels=3)
+    out = model(torch.randn(2, 3, 32, 32))
+    assert out.shape == (2, 10)
+    out.sum().backward()

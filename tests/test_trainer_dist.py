@@ -102,9 +102,10 @@ def _run_rank_p2p(rank, world, port, num_agg, q):
     from atomo_amd.parallel import Comm, PSTrainer
 
     comm = Comm(backend="gloo", device=torch.device("cpu"))
+    code = "qsgd" if num_agg == 2 else "svd"
     trainer = PSTrainer(
         model_name="LeNet",
-        codec=make_codec("svd", rank=3),
+        codec=make_codec(code, rank=3, quantization_level=4, bucket_size=256),
         comm=comm, lr=0.05, momentum=0.9, num_classes=10, in_channels=1,
         seed=7, device=torch.device("cpu"), comm_type="P2P",
         num_aggregate=num_agg,

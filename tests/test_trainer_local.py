@@ -148,3 +148,21 @@ def test_compress_false_svd_behaves_like_raw():
     train, _ = make_loaders("mnist", 16, 16, torch.device("cpu"), seed=9)
     losses = [trainer.train_step(x, y) for _, (x, y) in zip(range(20), train)]
     assert sum(losses[-5:]) / 5 < sum(losses[:5]) / 5
+
+
+def test_checkpoint_resume_with_svd(tmp_path):
+    trainer = _make_trainer(code="svd")
+    trainer.train_dir = str(tmp_path)
+    train, _ = make_loaders("mnist", 16, 16, torch.device("cpu"), seed=3)
+    it = iter(train)
+    for _ in range(3):
+        x, y = next(it)
+        trainer.train_step(x, y)
+    path = trainer.save_checkpoint()
+    trainer2 = _make_trainer(code="svd")
+    trainer2.load_checkpoint(path)
+    assert trainer2.step_num == 3
+    assert torch.allclose(trainer2.flat, trainer.flat)
+    x, y = next(it)
+    trainer2.train_step(x, y)  # resumed trainer keeps stepping
+    assert torch.isfinite(trainer2.flat).all()
