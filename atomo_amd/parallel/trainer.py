@@ -9,8 +9,10 @@ DistributedWorker.train, distributed_worker.py:166-262).  Per global step:
   2. comp    forward/backward on this rank's synthetic batch (skipped by a
              dedicated PS).
   3. encode  per-layer codec encode straight into the fixed wire bucket.
-  4. comm    ONE gather of wire buckets to rank 0 (or ONE reduce for the
-             raw codec — RCCL sums on the wire).
+  4. comm    ONE gather of wire buckets to rank 0 (ONE reduce for the raw
+             codec — RCCL sums on the wire; ``comm_type="P2P"`` switches to
+             arrival-order isend/irecv with decode-as-arrives and optional
+             first-K ``num_aggregate`` partial aggregation).
   5. decode  PS accumulates all workers' packets into the flat agg buffer
              (fused HIP kernels on GPU).
   6. apply   external-grad optimizer step with grad_scale = 1/num_workers,
