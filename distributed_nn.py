@@ -56,6 +56,10 @@ def main(argv=None):
                     "iters_per_sec": step / dt,
                 }
                 rec.update(trainer.timers.summary())
+                dev_bytes = trainer.wc.device_msg_bytes()
+                if dev_bytes > 0:
+                    # device-side Msg counter (async sampler); cumulative
+                    rec["msg_bytes_device_total"] = dev_bytes
                 print(json.dumps(rec), flush=True)
             if a.eval_freq and step % a.eval_freq == 0 and trainer.is_worker:
                 ev = trainer.evaluate(test_loader)
