@@ -67,6 +67,9 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
                         "default colocates a worker on rank 0")
     p.add_argument("--shrink-freq", type=int, default=50)
     p.add_argument("--checkpoint-freq", type=int, default=0)
+    p.add_argument("--resume", type=str, default=None, nargs="?", const="latest",
+                   help="checkpoint path to resume from, or 'latest' to pick "
+                        "the newest model_step_<N> in --train-dir")
     p.add_argument("--svd-backend", type=str, default="auto",
                    choices=["auto", "torch", "gram"])
     p.add_argument("--overlap", action="store_true", default=False,

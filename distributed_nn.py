@@ -23,6 +23,23 @@ from atomo_amd.data import make_loaders
 from atomo_amd.parallel import Comm, PSTrainer
 
 
+def resolve_checkpoint(resume, train_dir):
+    """'latest' -> newest model_step_<N> in train_dir; else the given path."""
+    import os
+    import re
+
+    if resume != "latest":
+        return resume if os.path.isfile(resume) else None
+    best, best_n = None, -1
+    if os.path.isdir(train_dir):
+        for fn in os.listdir(train_dir):
+            m = re.fullmatch(r"model_step_(\d+)", fn)
+            if m and int(m.group(1)) > best_n:
+                best_n = int(m.group(1))
+                best = os.path.join(train_dir, fn)
+    return best
+
+
 def main(argv=None):
     cfg = parse_args(argv)
     a = cfg.args
@@ -30,13 +47,24 @@ def main(argv=None):
     device = comm.device if cfg.device.type == "cuda" else cfg.device
     codec = cfg.build_codec()
     trainer = PSTrainer(codec=codec, comm=comm, device=device, **cfg.trainer_kwargs())
+    if a.resume:
+        path = resolve_checkpoint(a.resume, a.train_dir)
+        if path:
+            trainer.load_checkpoint(path)
+            if comm.rank == 0:
+                print(json.dumps({"log": "resume", "path": path,
+                                  "step": trainer.step_num}), flush=True)
+        elif comm.rank == 0:
+            print(json.dumps({"log": "resume", "path": None,
+                              "note": "no checkpoint found, fresh start"}),
+                  flush=True)
 
     train_loader, test_loader = make_loaders(
         a.dataset, a.batch_size, a.test_batch_size, device,
         seed=a.seed + comm.rank, root=a.data_root
     )
 
-    step, done = 0, False
+    step, done = trainer.step_num, False
     t0 = time.perf_counter()
     for epoch in range(a.epochs):
         if done:

@@ -66,3 +66,23 @@ def test_distributed_nn_torchrun_cpu():
     train = [r for r in recs if r.get("log") == "train"]
     assert train, out.stdout[-2000:]
     assert train[-1]["step"] == 4
+
+
+def test_resume_latest(tmp_path, capsys):
+    sys.path.insert(0, REPO)
+    import distributed_nn
+
+    args = [
+        "--network", "LeNet", "--dataset", "mnist", "--code", "sgd",
+        "--batch-size", "8", "--max-steps", "3", "--log-interval", "1",
+        "--checkpoint-freq", "3", "--train-dir", str(tmp_path), "--no-cuda",
+    ]
+    assert distributed_nn.main(args) == 0
+    capsys.readouterr()
+    # resume and run 3 more steps: checkpoint numbering continues
+    assert distributed_nn.main(args + ["--resume", "--max-steps", "6"]) == 0
+    out = capsys.readouterr().out
+    recs = [json.loads(l) for l in out.splitlines() if l.startswith("{")]
+    res = [r for r in recs if r.get("log") == "resume"]
+    assert res and res[0]["step"] == 3
+    assert os.path.isfile(os.path.join(str(tmp_path), "model_step_6"))
