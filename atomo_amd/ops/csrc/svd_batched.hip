@@ -220,7 +220,10 @@ void atomo_batched_sel_launch(const float* flat, float* wire,
                               const float* stage, const int64_t* desc,
                               const int32_t* work, int n_tiles, int sel_elems,
                               hipStream_t stream) {
-  const size_t lds = (size_t)sel_elems * 4 + 16;
+  // +16 for the cached-layer word; +128 because sel_rows' predicated
+  // (r < r_hat ? ...) reads may touch up to RCAP-1 floats past the last
+  // staged element (values never used, but keep them in-bounds)
+  const size_t lds = (size_t)sel_elems * 4 + 16 + 128;
   hipLaunchKernelGGL(batched_sel_kernel, dim3(grid_for_tiles(n_tiles)),
                      dim3(256), lds, stream, flat, wire, stage, desc, work,
                      n_tiles, sel_elems);
