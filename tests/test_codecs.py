@@ -179,3 +179,37 @@ def test_make_codec_dispatch():
     assert make_codec("qsgd", quantization_level=4).qlevel == 4
     with pytest.raises(ValueError):
         make_codec("nope")
+
+
+def test_svd_estimator_variance_identity(cpu_gen):
+    """The sampled estimator's variance matches theory:
+    E||X - A||_F^2 = sum_i (1/p_i - 1) s_i^2  (X = sum_{i in S} (s_i/p_i) u_i v_i^T,
+    independent Bernoulli inclusions — the paper's min-variance sampler)."""
+    torch.manual_seed(3)
+    a = torch.randn(40, 12)
+    u, s, vh = torch.linalg.svd(a, full_matrices=False)
+    rank = 3
+    probs = (rank * s / s.sum()).clamp(max=1.0)
+    expect_var = float(((1.0 / probs - 1.0) * s**2).sum())
+
+    codec = SVDCodec(rank=rank, generator=cpu_gen)
+    n = 2500
+    err2 = 0.0
+    for _ in range(n):
+        x = codec.decode(codec.encode(a))
+        err2 += float((x - a).norm() ** 2)
+    emp_var = err2 / n
+    assert abs(emp_var - expect_var) / expect_var < 0.1, (emp_var, expect_var)
+
+
+def test_svd_exact_when_budget_covers_spectrum(cpu_gen):
+    """Equal singular values with rank budget = n make every p_i = 1:
+    the 'sampled' code is the exact gradient."""
+    torch.manual_seed(4)
+    q, _ = torch.linalg.qr(torch.randn(32, 6))
+    w, _ = torch.linalg.qr(torch.randn(6, 6))
+    a = 2.5 * q @ w.t()  # all singular values = 2.5
+    codec = SVDCodec(rank=6, generator=cpu_gen)
+    for _ in range(5):
+        x = codec.decode(codec.encode(a))
+        assert torch.allclose(x, a, atol=1e-4), (x - a).abs().max()
