@@ -116,9 +116,11 @@ class BatchedSVDEncoder:
         specs: List[LayerSpec],
         device: torch.device,
         param_offsets: Optional[List[int]] = None,
+        rank: int = 0,
     ):
         self.codec = codec
         self.device = device
+        self.comm_rank = int(rank)
         self.specs = list(specs)
         self.param_offsets = param_offsets
         self._pool = ThreadPoolExecutor(max_workers=8)
@@ -269,7 +271,12 @@ class BatchedSVDEncoder:
                 self.grams_host = torch.zeros(gram_off, dtype=torch.float32, pin_memory=True)
                 self.sel_elems = sel_elems
                 self.used_words_dev = torch.zeros(1, dtype=torch.int64, device=dev)
-                self._seed = 0x9E3779B97F4A7C15
+                # rank-mixed: each worker's on-device Bernoulli sampler draws
+                # independent atoms (identical seeds collapse the
+                # 1/W compression-variance reduction of PS averaging)
+                self._seed = (
+                    0x9E3779B97F4A7C15 ^ (self.comm_rank * 0xD1B54A32D192ED03)
+                ) % (1 << 62)
             else:
                 self.use_kernels = False
 

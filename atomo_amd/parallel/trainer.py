@@ -76,10 +76,14 @@ class PSTrainer:
             # call take 600 ms.  Cap them.
             torch.set_num_threads(min(8, os.cpu_count() or 8))
         if seed is not None:
-            torch.manual_seed(seed)
+            # rank-decorrelated: host-side compression draws (sample_svd,
+            # QSGD dice) ride the global torch RNG — every worker needs its
+            # own stream.  Model-init consistency is guaranteed separately
+            # by the flat-param broadcast from rank 0 below.
+            torch.manual_seed(seed + comm.rank * 1000003)
         self.model = build_model(model_name, num_classes, in_channels).to(self.device)
         self.flat, self.params = flatten_params(self.model)
-        self.wc = WireCodec(codec, self.params, self.device)
+        self.wc = WireCodec(codec, self.params, self.device, rank=comm.rank)
         self.codec = codec
         self.loss_fn = nn.CrossEntropyLoss()
 
@@ -406,21 +410,17 @@ class PSTrainer:
         return path
 
     def load_checkpoint(self, path: str) -> None:
+        """In-place resume: ``load_state_dict(assign=False)`` copies every
+        tensor INTO the existing parameter views, so the flat buffer, the
+        grad views, the WireCodec/encoder and any registered overlap hooks
+        all stay valid — nothing is rebuilt.  (An earlier version rebuilt
+        the WireCodec here, which orphaned the backward-hook overlap state
+        and silently zeroed the SVD spectrum after resume.)"""
         ckpt = torch.load(path, map_location=self.device, weights_only=False)
-        self.model.load_state_dict(ckpt["model"])
-        # state_dict load replaced param storages; re-flatten into our buffer
-        self.flat, self.params = flatten_params(self.model)
-        self.wc = WireCodec(self.codec, self.params, self.device)
-        self.flat_grad = torch.zeros_like(self.flat)
-        off = 0
-        for p in self.params:
-            n = p.numel()
-            p.grad = self.flat_grad[off : off + n].view_as(p)
-            off += n
-        if self.wc.reducible:
-            self.wire = self.flat_grad
+        with torch.no_grad():
+            self.model.load_state_dict(ckpt["model"], assign=False)
+        self.flat_grad.zero_()
         self.step_num = ckpt["step"]
         self.lr = ckpt["lr"]
         if self.opt is not None and ckpt.get("optimizer") is not None:
-            self.opt.p = self.flat
             self.opt.load_state_dict(ckpt["optimizer"])

@@ -23,10 +23,22 @@ from ..codings.base import LayerSpec
 class WireCodec:
     """Binds a codec to a parameter list: layout, encode, decode."""
 
-    def __init__(self, codec: Codec, params: List[torch.Tensor], device: torch.device):
+    def __init__(
+        self,
+        codec: Codec,
+        params: List[torch.Tensor],
+        device: torch.device,
+        rank: int = 0,
+    ):
         self.codec = codec
         self.params = params
         self.device = device
+        # per-rank RNG decorrelation: every worker must make INDEPENDENT
+        # stochastic-rounding / atom-selection draws (the reference's MPI
+        # processes each have their own np.random state), otherwise the PS
+        # average keeps ~one worker's compression variance instead of 1/W.
+        self.rank = int(rank)
+        rank_mix = self.rank * 0x9E3779B97F4A7C15
         self.specs = codec.build_specs([list(p.shape) for p in params])
         self.total_words = sum(s.wire_words for s in self.specs)
         self._batched_encoder = None
@@ -46,13 +58,19 @@ class WireCodec:
             from .svd_encoder import BatchedSVDEncoder
 
             self._batched_encoder = BatchedSVDEncoder(
-                codec, self.specs, device, param_offsets=self.param_offsets
+                codec,
+                self.specs,
+                device,
+                param_offsets=self.param_offsets,
+                rank=self.rank,
             )
         # batched QSGD tables: one pack / unpack launch for the whole model
         self._qsgd_tables = None
         if device.type == "cuda" and isinstance(codec, QSGDCodec):
             from .. import ops
+            from ..ops import qsgd_ops
 
+            qsgd_ops.set_seed(12345 ^ rank_mix)
             if ops.have_ext():
                 desc, pack_work, unpack_work = [], [], []
                 for spec, p_off in zip(self.specs, self.param_offsets):
@@ -72,7 +90,7 @@ class WireCodec:
                         unpack_work, dtype=torch.int32, device=device
                     ),
                 }
-                self._qsgd_seed = 987654321
+                self._qsgd_seed = (987654321 ^ rank_mix) % (1 << 62)
         # scratch for layers whose 2-D fold is zero-padded (odd 1-D sizes)
         self._pad_scratch = {}
         if isinstance(codec, SVDCodec) and codec.compress:
@@ -142,6 +160,7 @@ class WireCodec:
 
         if self.device.type != "cuda":
             return False
+        self._overlap_handles = []
         if isinstance(self.codec, SVDCodec) and self._batched_encoder is not None:
             enc = self._batched_encoder
             enc.setup_solver_overlap(side_stream)
@@ -169,7 +188,9 @@ class WireCodec:
                 return hook
 
             for i, p in enumerate(params):
-                p.register_post_accumulate_grad_hook(make_hook(i))
+                self._overlap_handles.append(
+                    p.register_post_accumulate_grad_hook(make_hook(i))
+                )
             return True
         if isinstance(self.codec, QSGDCodec):
             from ..ops import qsgd_ops
@@ -196,9 +217,17 @@ class WireCodec:
                 return hook
 
             for i, p in enumerate(params):
-                p.register_post_accumulate_grad_hook(make_hook(i))
+                self._overlap_handles.append(
+                    p.register_post_accumulate_grad_hook(make_hook(i))
+                )
             return True
         return False
+
+    def remove_overlap(self) -> None:
+        """Unregister every backward hook installed by setup_overlap."""
+        for h in getattr(self, "_overlap_handles", []):
+            h.remove()
+        self._overlap_handles = []
 
     def arm_overlap(self) -> None:
         if self._batched_encoder is not None:

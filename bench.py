@@ -105,6 +105,7 @@ def main():
 
     sync()
     trainer.timers.reset()
+    trainer.wc.reset_device_msg_bytes()  # count timed steps only
     if a.phase_log:
         trainer.timers.sync_cuda = device.type == "cuda"
     t0 = time.perf_counter()
@@ -133,12 +134,10 @@ def main():
     images_per_sec = iters_per_sec * a.batch_size * n_workers
     msg_bytes = trainer.timers.scalars.get("msg_bytes", 0.0)
     steps_counted = max(1, trainer.timers.counts.get("msg_bytes", 1))
-    # device counter covers warmup + timed steps; the host scalar covers the
-    # timed steps of host-path layers — the two partition the layers
+    # both counters cover the timed steps only (device counter reset after
+    # warmup); host scalar and device counter partition the layers
     dev_bytes = trainer.wc.device_msg_bytes()
-    grad_mb_per_step = (
-        msg_bytes / steps_counted + dev_bytes / (a.steps + a.warmup)
-    ) / 1e6
+    grad_mb_per_step = (msg_bytes / steps_counted + dev_bytes / a.steps) / 1e6
 
     if comm.rank == 0:
         print(

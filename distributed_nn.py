@@ -84,6 +84,11 @@ def main(argv=None):
                     "iters_per_sec": step / dt,
                 }
                 rec.update(trainer.timers.summary())
+                # wire-budget overflow: steps where the Bernoulli sample drew
+                # more atoms than r_max and the tail was truncated (keeps the
+                # estimator's bias surface visible; see codings/svd.py)
+                if getattr(codec, "overflow_count", 0):
+                    rec["svd_overflow_count"] = codec.overflow_count
                 dev_bytes = trainer.wc.device_msg_bytes()
                 if dev_bytes > 0:
                     # device-side Msg counter (async sampler); cumulative

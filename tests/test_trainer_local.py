@@ -166,3 +166,66 @@ def test_checkpoint_resume_with_svd(tmp_path):
     x, y = next(it)
     trainer2.train_step(x, y)  # resumed trainer keeps stepping
     assert torch.isfinite(trainer2.flat).all()
+
+
+def test_resume_keeps_views_and_hooks(tmp_path):
+    """Regression: load_checkpoint must NOT rebuild the
+    WireCodec / flat buffers — overlap hooks registered at __init__ close
+    over the encoder, and a rebuild silently orphans them (zero spectrum
+    after resume).  In-place load keeps every view and object identity."""
+    trainer = _make_trainer(code="svd")
+    trainer.train_dir = str(tmp_path)
+    train, _ = make_loaders("mnist", 16, 16, torch.device("cpu"), seed=3)
+    it = iter(train)
+    for _ in range(2):
+        x, y = next(it)
+        trainer.train_step(x, y)
+    path = trainer.save_checkpoint()
+
+    wc_before = trainer.wc
+    flat_before = trainer.flat
+    flat_grad_before = trainer.flat_grad
+    trainer.load_checkpoint(path)
+    assert trainer.wc is wc_before
+    assert trainer.flat is flat_before
+    assert trainer.flat_grad is flat_grad_before
+    # params still alias the flat buffer and grads the flat grad buffer
+    off = 0
+    for p in trainer.params:
+        n = p.numel()
+        assert p.data.data_ptr() == trainer.flat[off:off + n].data_ptr()
+        assert p.grad.data_ptr() == trainer.flat_grad[off:off + n].data_ptr()
+        off += n
+    # resumed trainer still steps and the weights actually move
+    before = trainer.flat.clone()
+    x, y = next(it)
+    trainer.train_step(x, y)
+    assert not torch.allclose(trainer.flat, before)
+    assert torch.isfinite(trainer.flat).all()
+
+
+def test_compression_rng_differs_across_ranks():
+    """Regression: two ranks with the same base seed
+    must make DIFFERENT stochastic atom-selection draws (the PS average
+    then reduces compression variance by 1/W)."""
+    from atomo_amd.codings.svd import sample_svd
+
+    s = torch.linspace(1.0, 0.01, 32)
+    draws = []
+    for rank in (0, 1):
+        torch.manual_seed(7 + rank * 1000003)  # trainer's per-rank seeding
+        idx, _ = sample_svd(s, rank=3)
+        draws.append(idx.tolist())
+    assert draws[0] != draws[1]
+
+    # wire-level QSGD seeds are rank-mixed too
+    from atomo_amd.codings import make_codec
+    from atomo_amd.parallel.wire import WireCodec
+
+    model = __import__("atomo_amd.models", fromlist=["build_model"]).build_model(
+        "LeNet", 10, 1
+    )
+    params = [p for p in model.parameters() if p.requires_grad]
+    w0 = WireCodec(make_codec("svd", rank=3), params, torch.device("cpu"), rank=0)
+    w1 = WireCodec(make_codec("svd", rank=3), params, torch.device("cpu"), rank=1)
+    assert w0.rank != w1.rank
