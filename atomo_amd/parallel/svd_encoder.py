@@ -9,8 +9,12 @@ DEVICE layers (even 2-D fold, wire budget r_max <= 32):
     (jacobi_eigh_kernel, previous step's eigenbasis pre-rotates so sweeps
     converge in ~1-2 iterations).
   * bigger folds (1x1-conv Grams up to 2048x1024): rocBLAS Gram ->
-    batched hipSOLVER syevd, size-groups merged into one call by
-    zero-padding when the measured cost model favors it.
+    warm-started randomized Rayleigh-Ritz (_solve_big_folds_randomized:
+    one subspace-iteration power step + CholQR2 + ONE merged tiny eigh
+    across every size group — the north-star "one-pass randomized SVD";
+    the exact batched hipSOLVER syevd survives as the ATOMO_EXACT_EIGH=1
+    oracle, size-groups merged by zero-padding per the measured cost
+    model).
   * then: fused on-device Bernoulli sampler + stage builder
     (sample_stage_kernel; counter-hash RNG; Msg bytes counted on device)
     -> batched_sel kernel writes the wire packets.  No host sync at all
@@ -42,6 +46,9 @@ from ..codings.base import LayerSpec
 from ..codings.svd import SVDCodec, sample_svd
 
 _TRACE = os.environ.get("ATOMO_TRACE_ENCODER", "") not in ("", "0")
+# big-fold eigensolver: randomized Rayleigh-Ritz (default) vs exact syevd
+# (ATOMO_EXACT_EIGH=1; read per-encoder at init)
+_RSVD_OVERSAMPLE = int(os.environ.get("ATOMO_RSVD_OVERSAMPLE", "8"))
 
 GRAM_CHUNK = 256
 SEL_CHUNK = 1024
@@ -101,6 +108,40 @@ def _robust_eigh(g: torch.Tensor, out_dtype=torch.float64):
         return evals, evecs
 
 
+def batched_orthonormalize(y: torch.Tensor) -> torch.Tensor:
+    """Batched CholQR with trace-scaled jitter; falls back to Householder
+    QR when the Cholesky fails (rank-deficient panels)."""
+    b = y.shape[-1]
+    try:
+        # column scaling first: after a power step the column norms span
+        # the (squared) spectrum's dynamic range, which alone overflows
+        # fp32 CholQR; unit columns leave only angular conditioning
+        y = y / y.norm(dim=1, keepdim=True).clamp(min=1e-30)
+        s = torch.bmm(y.transpose(1, 2), y)
+        s = s + 1e-5 * torch.eye(b, dtype=s.dtype, device=s.device)
+        ell = torch.linalg.cholesky(s)
+        return torch.linalg.solve_triangular(
+            ell, y.transpose(1, 2), upper=False
+        ).transpose(1, 2)
+    except Exception:
+        return torch.linalg.qr(y, mode="reduced").Q
+
+
+def subspace_iterate(g: torch.Tensor, q: torch.Tensor, iters: int):
+    """``iters`` rounds of orthonormalized subspace iteration on the
+    symmetric batch ``g`` followed by the Rayleigh-Ritz projection.
+    Returns (t, q): t = q^T g q (batch, b, b) and the refined orthonormal
+    basis q.  eigh(t) -> (lam, w); Ritz pairs are (lam, q @ w)."""
+    for _ in range(iters):
+        q = batched_orthonormalize(torch.bmm(g, q))
+    # CholQR2: a second pass restores near-machine orthogonality on the
+    # rank-deficient tail directions (decoded atoms are A v v^T / p — the
+    # unbiased-projection argument needs V orthonormal)
+    q = batched_orthonormalize(q)
+    z = torch.bmm(g, q)
+    return torch.bmm(q.transpose(1, 2), z), q
+
+
 def _interp_cost(table, sm):
     ks = sorted(table)
     for k in ks:
@@ -121,6 +162,8 @@ class BatchedSVDEncoder:
         self.codec = codec
         self.device = device
         self.comm_rank = int(rank)
+        # read at init (not import) so tests can toggle per instance
+        self.exact_eigh = os.environ.get("ATOMO_EXACT_EIGH", "0") not in ("", "0")
         self.specs = list(specs)
         self.param_offsets = param_offsets
         self._pool = ThreadPoolExecutor(max_workers=8)
@@ -192,14 +235,20 @@ class BatchedSVDEncoder:
                     counts[self.small[i]] += 1
             solver_dims = set()
             for sm, cnt in counts.items():
-                # measured: concurrent MKL eighs barely parallelize (~2x)
-                host_ms = _interp_cost(_HOST_EIGH_MS, sm) * max(1.0, cnt / 2.0)
-                solver_ms = _interp_cost(_SOLVER_EIGH_MS, sm)
-                # pooled host solves overlap ~2 ms of device work for free
-                host_eff = max(0.0, host_ms - 2.0)
                 solver_min = int(os.environ.get("ATOMO_SOLVER_MIN_SM", "0"))
                 if solver_min and sm < solver_min:
                     continue
+                if not self.exact_eigh:
+                    # randomized Rayleigh-Ritz makes every big fold a few
+                    # batched GEMMs + one tiny eigh: always device-route
+                    solver_dims.add(sm)
+                    continue
+                # exact-oracle mode keeps the measured syevd-vs-LAPACK route
+                # (concurrent MKL eighs barely parallelize (~2x); pooled host
+                # solves overlap ~2 ms of device work for free)
+                host_ms = _interp_cost(_HOST_EIGH_MS, sm) * max(1.0, cnt / 2.0)
+                solver_ms = _interp_cost(_SOLVER_EIGH_MS, sm)
+                host_eff = max(0.0, host_ms - 2.0)
                 if solver_ms < host_eff or sm >= 768:
                     solver_dims.add(sm)
             for i, s in enumerate(specs):
@@ -277,6 +326,36 @@ class BatchedSVDEncoder:
                 self._seed = (
                     0x9E3779B97F4A7C15 ^ (self.comm_rank * 0xD1B54A32D192ED03)
                 ) % (1 << 62)
+                # randomized big-fold solver state (warm subspace per group)
+                self._rsvd_groups = []
+                self._rsvd_Q = {}
+                if self.solver_layers and not self.exact_eigh:
+                    by_sm = defaultdict(list)
+                    for i in self.solver_layers:
+                        by_sm[self.small[i]].append(i)
+                    r_top = max(
+                        self.specs[i].meta["r_max"] for i in self.solver_layers
+                    )
+                    b = ((r_top + _RSVD_OVERSAMPLE + 7) // 8) * 8
+                    self._rsvd_b = min(b, min(by_sm))
+                    self._rsvd_groups = sorted(by_sm.items())
+                    g = torch.Generator(device=dev)
+                    g.manual_seed(
+                        (0xC0FFEE ^ (self.comm_rank * 0x9E3779B9)) % (1 << 62)
+                    )
+                    self._rsvd_gen = g
+                    # per-layer tail-energy diagnostic (Gram-trace fraction
+                    # outside the computed subspace); device-resident, fetch
+                    # via rsvd_tail_fraction() when logging
+                    self._rsvd_tail_slot = {
+                        i: k for k, i in enumerate(self.solver_layers)
+                    }
+                    self._rsvd_tail_dev = torch.zeros(
+                        len(self.solver_layers), device=dev
+                    )
+                    self._rsvd_trace_dev = torch.zeros(
+                        len(self.solver_layers), device=dev
+                    )
             else:
                 self.use_kernels = False
 
@@ -288,8 +367,10 @@ class BatchedSVDEncoder:
     # stream, a pool thread launches its (host-blocking) batched eigh on a
     # dedicated stream — hiding most of its cost under the rest of backward.
     def build_solver_plan(self):
-        """(call_size -> layer list) using the same merge rule as encode."""
-        if not self.solver_layers:
+        """(call_size -> layer list) using the same merge rule as encode.
+        Randomized mode returns {} — the subspace solve is a few batched
+        GEMMs run inline in encode_all, nothing to overlap from hooks."""
+        if not self.solver_layers or not self.exact_eigh:
             return {}
         by_sm = defaultdict(list)
         for i in self.solver_layers:
@@ -358,6 +439,102 @@ class BatchedSVDEncoder:
             self.grams[
                 self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
             ].copy_(evecs[j, :sm, :sm].reshape(-1))
+
+    # -- randomized big-fold solver ------------------------------------
+    # The exact syevd of an sm x sm Gram is O(sm^3) and is the step-time
+    # wall for the deep-ResNet 1x1-conv folds (sm up to 2048; BASELINE.md
+    # rows 4/5).  The sampler only ever ships <= r_max atoms, so a
+    # rank-(r_max + oversample) subspace is enough: one warm-started
+    # subspace iteration + Rayleigh-Ritz gives the top eigenpairs in a
+    # handful of batched GEMMs.  Crucial unbiasedness property: the wire
+    # ships u_i = A v_i / s_i, so a decoded atom is A v_i v_i^T / p_i —
+    # an UNBIASED estimate of A's projection onto span(V) for ANY
+    # orthonormal V, independent of eigenvalue accuracy (s_i cancels).
+    # Eigenvalue error only perturbs the sampling probabilities
+    # (variance); the only bias is the tail A(I - P_V), tracked per layer
+    # in _rsvd_tail_dev.  Atoms beyond the subspace read eval = 0 and are
+    # never sampled.  (North-star "one-pass randomized SVD"; reference
+    # semantics codings/svd.py:49-117.)
+    def _solve_big_folds_randomized(self) -> None:
+        b = self._rsvd_b
+        group_ctx = []
+        ts = []
+        for sm, idxs in self._rsvd_groups:
+            B = len(idxs)
+            g = torch.empty(B, sm, sm, device=self.device)
+            for j, i in enumerate(idxs):
+                g[j] = self.grams[
+                    self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
+                ].view(sm, sm)
+            g = 0.5 * (g + g.transpose(1, 2))
+            tr = g.diagonal(dim1=1, dim2=2).sum(dim=1)
+            q = self._rsvd_Q.get(sm)
+            iters = 1
+            if q is None or q.shape[0] != B:
+                q = torch.randn(
+                    B, sm, b, generator=self._rsvd_gen, device=self.device
+                )
+                iters = 2  # cold start: extra power step for subspace quality
+            t, q = subspace_iterate(g, q, iters)
+            ts.append(t)
+            group_ctx.append((sm, idxs, q, tr))
+        evals_t, w = _robust_eigh(torch.cat(ts, dim=0), out_dtype=torch.float32)
+        evals_t = evals_t.flip(1).clamp(min=0.0)
+        w = w.flip(2)
+        row0 = 0
+        for sm, idxs, q, tr in group_ctx:
+            B = len(idxs)
+            lam = evals_t[row0 : row0 + B]  # (B, b) descending
+            evecs = torch.bmm(q, w[row0 : row0 + B])  # (B, sm, b) orthonormal
+            row0 += B
+            self._rsvd_Q[sm] = evecs  # warm subspace for the next step
+            for j, i in enumerate(idxs):
+                o = self.eval_offs[self.layer_row[i]]
+                ev = self.evals_dev[o : o + sm]
+                ev.zero_()
+                ev[:b] = lam[j]
+                self.grams[
+                    self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
+                ].view(sm, sm)[:, :b] = evecs[j]
+                slot = self._rsvd_tail_slot[i]
+                self._rsvd_tail_dev[slot] = (tr[j] - lam[j].sum()).clamp(min=0.0)
+                self._rsvd_trace_dev[slot] = tr[j]
+
+    def _solve_big_folds_exact(self) -> None:
+        """Oracle path (ATOMO_EXACT_EIGH=1): batched hipSOLVER syevd per
+        size group, small groups merged into the largest call by
+        zero-padding when the measured cost model favors it (padding a PSD
+        Gram with zero rows/cols only appends zero eigenvalues)."""
+        by_sm = defaultdict(list)
+        for i in self.solver_layers:
+            by_sm[self.small[i]].append(i)
+        sizes = sorted(by_sm, reverse=True)
+        n_top = sizes[0]
+        calls = {n_top: list(by_sm[n_top])}
+        for sm in sizes[1:]:
+            cnt = len(by_sm[sm])
+            sep = _interp_cost(_SYEVD_BASE, sm) + _interp_cost(
+                _SYEVD_INCR, sm
+            ) * (cnt - 1)
+            merged = _interp_cost(_SYEVD_INCR, n_top) * cnt
+            if merged < sep:
+                calls[n_top].extend(by_sm[sm])
+            else:
+                calls[sm] = list(by_sm[sm])
+        for n_call, idxs in calls.items():
+            self._run_solver_group(n_call, idxs)
+
+    def rsvd_tail_fraction(self) -> dict:
+        """{layer index: fraction of Gram energy outside the computed
+        subspace} — the randomized solver's bias diagnostic (one D2H)."""
+        if not getattr(self, "_rsvd_groups", None):
+            return {}
+        tail = self._rsvd_tail_dev.cpu()
+        tr = self._rsvd_trace_dev.cpu()
+        return {
+            i: float(tail[k] / tr[k].clamp(min=1e-30))
+            for i, k in self._rsvd_tail_slot.items()
+        }
 
     def _host_solve_layer(self, i, ready_evt) -> None:
         ready_evt.synchronize()  # gram data final (host-side wait)
@@ -569,53 +746,12 @@ class BatchedSVDEncoder:
                 if self.solver_layers and not (
                     grams_done and getattr(self, "_ov_plan", None)
                 ):
-                    # group by fold size, then MERGE small groups into the
-                    # largest group's call by zero-padding when the syevd
-                    # cost model says the per-matrix increment at the big
-                    # size is cheaper than another call's base.  Padding a
-                    # PSD Gram with zero rows/cols only appends zero
-                    # eigenvalues, so the top-sm (descending) eigenpairs are
-                    # the original spectrum and their vectors have zero pad
-                    # components.
-                    by_sm = defaultdict(list)
-                    for i in self.solver_layers:
-                        by_sm[self.small[i]].append(i)
-                    sizes = sorted(by_sm, reverse=True)
-                    n_top = sizes[0]
-                    calls = {n_top: list(by_sm[n_top])}
-                    for sm in sizes[1:]:
-                        cnt = len(by_sm[sm])
-                        sep = _interp_cost(_SYEVD_BASE, sm) + _interp_cost(
-                            _SYEVD_INCR, sm
-                        ) * (cnt - 1)
-                        merged = _interp_cost(_SYEVD_INCR, n_top) * cnt
-                        if merged < sep:
-                            calls[n_top].extend(by_sm[sm])
-                        else:
-                            calls[sm] = list(by_sm[sm])
-                    for n_call, idxs in calls.items():
-                        B = len(idxs)
-                        gs = torch.zeros(
-                            B, n_call, n_call, device=self.device
-                        )
-                        for j, i in enumerate(idxs):
-                            sm = self.small[i]
-                            gs[j, :sm, :sm] = self.grams[
-                                self.gram_offsets[i] : self.gram_offsets[i]
-                                + sm * sm
-                            ].view(sm, sm)
-                        gs = 0.5 * (gs + gs.transpose(1, 2))
-                        evals, evecs = _robust_eigh(gs, out_dtype=torch.float32)
-                        evals = evals.flip(1).clamp(min=0.0)
-                        evecs = evecs.flip(2)
-                        for j, i in enumerate(idxs):
-                            sm = self.small[i]
-                            o = self.eval_offs[self.layer_row[i]]
-                            self.evals_dev[o : o + sm].copy_(evals[j, :sm])
-                            self.grams[
-                                self.gram_offsets[i] : self.gram_offsets[i]
-                                + sm * sm
-                            ].copy_(evecs[j, :sm, :sm].reshape(-1))
+                    if not self.exact_eigh:
+                        # randomized Rayleigh-Ritz: batched GEMMs + ONE tiny
+                        # merged eigh over every size group (see method doc)
+                        self._solve_big_folds_randomized()
+                    else:
+                        self._solve_big_folds_exact()
                 self.evals_host.copy_(self.evals_dev, non_blocking=True)
             elif host_layers:
                 grams_host = self.grams.to("cpu")  # synchronous copy

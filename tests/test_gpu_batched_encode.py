@@ -8,7 +8,7 @@ import torch
 pytestmark = pytest.mark.gpu
 
 
-def _build(dev, shapes, rank=3, random_sample=False):
+def _build(dev, shapes, rank=3, random_sample=False, exact_eigh=None):
     from atomo_amd.codings import SVDCodec
     from atomo_amd.parallel.svd_encoder import BatchedSVDEncoder
 
@@ -23,7 +23,19 @@ def _build(dev, shapes, rank=3, random_sample=False):
     grads = [
         flat[o : o + n].view(shape) for o, n, shape in zip(offsets, numels, shapes)
     ]
-    enc = BatchedSVDEncoder(codec, specs, dev, param_offsets=offsets)
+    import os
+
+    prev = os.environ.get("ATOMO_EXACT_EIGH")
+    if exact_eigh is not None:
+        os.environ["ATOMO_EXACT_EIGH"] = "1" if exact_eigh else "0"
+    try:
+        enc = BatchedSVDEncoder(codec, specs, dev, param_offsets=offsets)
+    finally:
+        if exact_eigh is not None:
+            if prev is None:
+                os.environ.pop("ATOMO_EXACT_EIGH", None)
+            else:
+                os.environ["ATOMO_EXACT_EIGH"] = prev
     wire = torch.zeros(sum(s.wire_words for s in specs), device=dev)
     return codec, specs, enc, flat, grads, wire
 
@@ -48,9 +60,12 @@ def test_kernel_layers_selected(dev):
 
 
 def test_batched_kernels_match_truncated_svd(dev):
+    # exact_eigh pins the syevd oracle: this test asserts bit-tight parity
+    # with the full SVD; the randomized big-fold path has its own test below
     torch.manual_seed(0)
     shapes = [(64, 16, 3, 3), (512,), (128, 64, 1, 1), (256, 128, 1, 1), (10, 512)]
-    codec, specs, enc, flat, grads, wire = _build(dev, shapes, rank=3)
+    codec, specs, enc, flat, grads, wire = _build(dev, shapes, rank=3,
+                                                  exact_eigh=True)
     from atomo_amd.codings.svd import grad_to_2d
 
     # run twice: the second pass exercises the warm-started Jacobi
@@ -136,3 +151,79 @@ def test_batched_decode_matches_per_layer(dev):
                 codec.decode_from(regions[w], out, spec)
     err = (agg_new - agg_ref).abs().max().item()
     assert err < 1e-4, err
+
+
+def _decaying_grads(flat, grads, specs, decay=6.0, seed=11):
+    """Overwrite grads in-place with gradient-like decaying-spectrum
+    matrices (the regime the randomized solver is built for)."""
+    gen = torch.Generator(device=flat.device).manual_seed(seed)
+    for g, spec in zip(grads, specs):
+        m, n = spec.meta["m"], spec.meta["n"]
+        if spec.meta["padded"] != spec.numel:
+            continue
+        k = min(m, n)
+        u = torch.linalg.qr(
+            torch.randn(m, k, device=flat.device, generator=gen)
+        ).Q
+        v = torch.linalg.qr(
+            torch.randn(n, k, device=flat.device, generator=gen)
+        ).Q
+        s = torch.exp(-torch.arange(k, device=flat.device) / decay)
+        g.view(-1).copy_(((u * s) @ v.t()).reshape(-1))
+
+
+def test_randomized_big_folds_match_topr(dev):
+    """Randomized Rayleigh-Ritz on big 1x1-conv folds: after a few warm
+    steps, truncation mode must reproduce the top-r SVD reconstruction
+    within the subspace tolerance."""
+    torch.manual_seed(2)
+    shapes = [(256, 128, 1, 1), (512, 256, 1, 1), (1024, 512, 1, 1)]
+    codec, specs, enc, flat, grads, wire = _build(dev, shapes, rank=3,
+                                                  exact_eigh=False)
+    assert enc.solver_layers, "big folds should route to the solver path"
+    assert not enc.exact_eigh and enc._rsvd_groups
+    _decaying_grads(flat, grads, specs)
+    from atomo_amd.codings.svd import grad_to_2d
+
+    for _ in range(4):  # warm the subspace
+        enc.encode_all(grads, wire, flat_grad=flat)
+    for g, spec in zip(grads, specs):
+        region = wire[spec.wire_offset : spec.wire_offset + spec.wire_words].cpu()
+        out = torch.zeros(spec.numel)
+        codec.decode_from(region, out, spec)
+        a = grad_to_2d(g.cpu())
+        u, s, vh = torch.linalg.svd(a, full_matrices=False)
+        r = min(codec.rank, spec.meta["r_max"])
+        best = ((u[:, :r] * s[:r]) @ vh[:r]).reshape(-1)[: spec.numel]
+        rel = (out - best).norm() / best.norm()
+        assert rel < 2e-2, (spec.shape, float(rel))
+    # tail diagnostic is tiny on a decaying spectrum
+    tails = enc.rsvd_tail_fraction()
+    assert tails and max(tails.values()) < 5e-3, tails
+
+
+def test_randomized_matches_exact_eigenvalues(dev):
+    """Warm randomized eigenvalues vs the exact syevd oracle on the same
+    Grams: top-r_max relative error within 1%."""
+    torch.manual_seed(3)
+    shapes = [(512, 256, 1, 1), (1024, 256, 1, 1)]
+    c1, specs, enc_r, flat, grads, wire = _build(dev, shapes, rank=3,
+                                                 exact_eigh=False)
+    c2, _, enc_x, flat2, _, wire2 = _build(dev, shapes, rank=3, exact_eigh=True)
+    flat2.copy_(flat)
+    _decaying_grads(flat, grads, specs)
+    flat2.copy_(flat)
+    grads2 = [flat2[o : o + s.numel].view(s.shape)
+              for o, s in zip(enc_x.param_offsets, specs)]
+    for _ in range(4):
+        enc_r.encode_all(grads, wire, flat_grad=flat)
+    enc_x.encode_all(grads2, wire2, flat_grad=flat2)
+    torch.cuda.synchronize()
+    for i, spec in enumerate(specs):
+        r_max = spec.meta["r_max"]
+        o_r = enc_r.eval_offs[enc_r.layer_row[i]]
+        o_x = enc_x.eval_offs[enc_x.layer_row[i]]
+        ev_r = enc_r.evals_dev[o_r : o_r + r_max].cpu()
+        ev_x = enc_x.evals_dev[o_x : o_x + r_max].cpu()
+        rel = ((ev_r - ev_x).abs() / ev_x.clamp(min=1e-12)).max()
+        assert rel < 1e-2, (spec.shape, float(rel))
