@@ -337,8 +337,8 @@ class BatchedSVDEncoder:
                         self.specs[i].meta["r_max"] for i in self.solver_layers
                     )
                     b = ((r_top + _RSVD_OVERSAMPLE + 7) // 8) * 8
-                    self._rsvd_b = min(b, min(by_sm))
-                    self._rsvd_groups = sorted(by_sm.items())
+                    b = min(b, min(by_sm))
+                    self._rsvd_b = b
                     g = torch.Generator(device=dev)
                     g.manual_seed(
                         (0xC0FFEE ^ (self.comm_rank * 0x9E3779B9)) % (1 << 62)
@@ -356,6 +356,36 @@ class BatchedSVDEncoder:
                     self._rsvd_trace_dev = torch.zeros(
                         len(self.solver_layers), device=dev
                     )
+                    # precomputed index tensors: ONE gather / scatter kernel
+                    # per group instead of O(layers) small slice copies (a
+                    # 97-layer python loop costs milliseconds in launches)
+                    for sm, idxs in sorted(by_sm.items()):
+                        g0 = torch.arange(sm * sm, dtype=torch.int64)
+                        gather = torch.cat(
+                            [g0 + self.gram_offsets[i] for i in idxs]
+                        ).to(dev)
+                        sc0 = (
+                            torch.arange(sm, dtype=torch.int64).view(-1, 1) * sm
+                            + torch.arange(b, dtype=torch.int64).view(1, -1)
+                        ).reshape(-1)
+                        scatter = torch.cat(
+                            [sc0 + self.gram_offsets[i] for i in idxs]
+                        ).to(dev)
+                        ev0 = torch.arange(b, dtype=torch.int64)
+                        evi = torch.cat(
+                            [
+                                ev0 + self.eval_offs[self.layer_row[i]]
+                                for i in idxs
+                            ]
+                        ).to(dev)
+                        tails = torch.tensor(
+                            [self._rsvd_tail_slot[i] for i in idxs],
+                            dtype=torch.int64,
+                            device=dev,
+                        )
+                        self._rsvd_groups.append(
+                            (sm, idxs, gather, scatter, evi, tails)
+                        )
             else:
                 self.use_kernels = False
 
@@ -459,13 +489,9 @@ class BatchedSVDEncoder:
         b = self._rsvd_b
         group_ctx = []
         ts = []
-        for sm, idxs in self._rsvd_groups:
+        for sm, idxs, gather, scatter, evi, tails in self._rsvd_groups:
             B = len(idxs)
-            g = torch.empty(B, sm, sm, device=self.device)
-            for j, i in enumerate(idxs):
-                g[j] = self.grams[
-                    self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
-                ].view(sm, sm)
+            g = self.grams.index_select(0, gather).view(B, sm, sm)
             g = 0.5 * (g + g.transpose(1, 2))
             tr = g.diagonal(dim1=1, dim2=2).sum(dim=1)
             q = self._rsvd_Q.get(sm)
@@ -477,28 +503,25 @@ class BatchedSVDEncoder:
                 iters = 2  # cold start: extra power step for subspace quality
             t, q = subspace_iterate(g, q, iters)
             ts.append(t)
-            group_ctx.append((sm, idxs, q, tr))
+            group_ctx.append((sm, q, tr, scatter, evi, tails))
         evals_t, w = _robust_eigh(torch.cat(ts, dim=0), out_dtype=torch.float32)
         evals_t = evals_t.flip(1).clamp(min=0.0)
         w = w.flip(2)
         row0 = 0
-        for sm, idxs, q, tr in group_ctx:
-            B = len(idxs)
+        for sm, q, tr, scatter, evi, tails in group_ctx:
+            B = q.shape[0]
             lam = evals_t[row0 : row0 + B]  # (B, b) descending
             evecs = torch.bmm(q, w[row0 : row0 + B])  # (B, sm, b) orthonormal
             row0 += B
             self._rsvd_Q[sm] = evecs  # warm subspace for the next step
-            for j, i in enumerate(idxs):
-                o = self.eval_offs[self.layer_row[i]]
-                ev = self.evals_dev[o : o + sm]
-                ev.zero_()
-                ev[:b] = lam[j]
-                self.grams[
-                    self.gram_offsets[i] : self.gram_offsets[i] + sm * sm
-                ].view(sm, sm)[:, :b] = evecs[j]
-                slot = self._rsvd_tail_slot[i]
-                self._rsvd_tail_dev[slot] = (tr[j] - lam[j].sum()).clamp(min=0.0)
-                self._rsvd_trace_dev[slot] = tr[j]
+            # single-kernel writebacks (evals tail slots stay zero from
+            # init -> atoms beyond the subspace are never sampled)
+            self.grams.index_copy_(0, scatter, evecs.reshape(-1))
+            self.evals_dev.index_copy_(0, evi, lam.reshape(-1))
+            self._rsvd_tail_dev.index_copy_(
+                0, tails, (tr - lam.sum(dim=1)).clamp(min=0.0)
+            )
+            self._rsvd_trace_dev.index_copy_(0, tails, tr)
 
     def _solve_big_folds_exact(self) -> None:
         """Oracle path (ATOMO_EXACT_EIGH=1): batched hipSOLVER syevd per
@@ -746,12 +769,14 @@ class BatchedSVDEncoder:
                 if self.solver_layers and not (
                     grams_done and getattr(self, "_ov_plan", None)
                 ):
+                    mark("A2 jacobi")
                     if not self.exact_eigh:
                         # randomized Rayleigh-Ritz: batched GEMMs + ONE tiny
                         # merged eigh over every size group (see method doc)
                         self._solve_big_folds_randomized()
                     else:
                         self._solve_big_folds_exact()
+                    mark("A3 big-fold solve")
                 self.evals_host.copy_(self.evals_dev, non_blocking=True)
             elif host_layers:
                 grams_host = self.grams.to("cpu")  # synchronous copy
