@@ -217,6 +217,9 @@ class BatchedSVDEncoder:
         self.kernel_rows: List[int] = []  # desc row -> layer index
         self.big_gram_layers: List[int] = []  # device rows needing rocBLAS gram
         self.use_kernels = False
+        self.hook_layers = set()
+        self.hook_mode = "all"
+        self._j64_zero_idx = None
         if device.type == "cuda" and param_offsets is not None:
             from .. import ops
 
@@ -310,6 +313,31 @@ class BatchedSVDEncoder:
                 self.vwarm = torch.zeros(max(1, vw), dtype=torch.float32, device=dev)
                 self.vwarm_offs = torch.tensor(vw_offs, dtype=torch.int64, device=dev)
                 self._warm = False
+                # selective-overlap support: backward hooks cover only the
+                # big-fold (rocBLAS mm) and host layers; j64 grams come from
+                # ONE batched_gram launch post-backward — hundreds of
+                # per-layer Python hooks during backward cost more host time
+                # than the single kernel (ResNet-152: 467 hooks -> 97).
+                # batched_gram accumulates atomically, so the j64 slots are
+                # zeroed selectively via this index when hooks own the rest.
+                self.hook_layers = set(self.big_gram_layers) | {
+                    i for i in range(len(specs)) if i not in self.kernel_set
+                }
+                self.hook_mode = "all"  # set by setup_overlap
+                j64_ids = [row_to_layer[r] for r in rows_j64]
+                self._j64_zero_idx = (
+                    torch.cat(
+                        [
+                            torch.arange(
+                                self.small[i] ** 2, dtype=torch.int64
+                            )
+                            + self.gram_offsets[i]
+                            for i in j64_ids
+                        ]
+                    ).to(dev)
+                    if j64_ids
+                    else None
+                )
                 self.eval_offs_dev = torch.tensor(eval_offs, dtype=torch.int64, device=dev)
                 self.eval_offs = eval_offs
                 self.evals_dev = torch.zeros(max(1, ev_off), dtype=torch.float32, device=dev)
@@ -718,6 +746,16 @@ class BatchedSVDEncoder:
             e = ext()
             if not grams_done:
                 self.grams.zero_()
+                if self.gram_work.shape[0]:
+                    e.batched_gram(
+                        flat_grad, self.grams, self.desc, self.gram_work,
+                        self.gram_work.shape[0],
+                    )
+            elif self.hook_mode == "big":
+                # hooks covered only big-fold/host grams; j64 grams run here
+                # as one launch (selective zero first: atomic accumulation)
+                if self._j64_zero_idx is not None:
+                    self.grams.index_fill_(0, self._j64_zero_idx, 0.0)
                 if self.gram_work.shape[0]:
                     e.batched_gram(
                         flat_grad, self.grams, self.desc, self.gram_work,

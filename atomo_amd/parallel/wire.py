@@ -162,8 +162,25 @@ class WireCodec:
             return False
         self._overlap_handles = []
         if isinstance(self.codec, SVDCodec) and self._batched_encoder is not None:
+            import os
+
             enc = self._batched_encoder
             enc.setup_solver_overlap(side_stream)
+            # "big": hook only big-fold + host layers; j64 grams run as one
+            # batched_gram launch post-backward.  "all": hook every layer
+            # (the round-1 behavior).  Default "auto": big when big folds
+            # exist (deep ResNets — hundreds of hooks cost more host time
+            # than one kernel), else all (small models, where per-layer
+            # grams fully hide under backward).
+            mode = os.environ.get("ATOMO_OVERLAP_MODE", "auto")
+            if mode == "auto":
+                mode = "big" if enc.hook_layers else "all"
+            enc.hook_mode = mode
+            hook_set = (
+                enc.hook_layers
+                if enc.hook_mode == "big"
+                else set(range(len(self.specs)))
+            )
 
             def make_hook(i):
                 spec = self.specs[i]
@@ -188,9 +205,10 @@ class WireCodec:
                 return hook
 
             for i, p in enumerate(params):
-                self._overlap_handles.append(
-                    p.register_post_accumulate_grad_hook(make_hook(i))
-                )
+                if i in hook_set:
+                    self._overlap_handles.append(
+                        p.register_post_accumulate_grad_hook(make_hook(i))
+                    )
             return True
         if isinstance(self.codec, QSGDCodec):
             from ..ops import qsgd_ops

@@ -205,27 +205,23 @@ def test_randomized_big_folds_match_topr(dev):
 
 
 def test_randomized_matches_exact_eigenvalues(dev):
-    """Warm randomized eigenvalues vs the exact syevd oracle on the same
-    Grams: top-r_max relative error within 1%."""
+    """Warm randomized eigenvalues vs the exact spectrum (CPU SVD oracle)
+    on the same gradients: top-r_max relative error within 1%."""
     torch.manual_seed(3)
     shapes = [(512, 256, 1, 1), (1024, 256, 1, 1)]
     c1, specs, enc_r, flat, grads, wire = _build(dev, shapes, rank=3,
                                                  exact_eigh=False)
-    c2, _, enc_x, flat2, _, wire2 = _build(dev, shapes, rank=3, exact_eigh=True)
-    flat2.copy_(flat)
     _decaying_grads(flat, grads, specs)
-    flat2.copy_(flat)
-    grads2 = [flat2[o : o + s.numel].view(s.shape)
-              for o, s in zip(enc_x.param_offsets, specs)]
     for _ in range(4):
         enc_r.encode_all(grads, wire, flat_grad=flat)
-    enc_x.encode_all(grads2, wire2, flat_grad=flat2)
     torch.cuda.synchronize()
+    from atomo_amd.codings.svd import grad_to_2d
+
     for i, spec in enumerate(specs):
         r_max = spec.meta["r_max"]
         o_r = enc_r.eval_offs[enc_r.layer_row[i]]
-        o_x = enc_x.eval_offs[enc_x.layer_row[i]]
         ev_r = enc_r.evals_dev[o_r : o_r + r_max].cpu()
-        ev_x = enc_x.evals_dev[o_x : o_x + r_max].cpu()
+        s = torch.linalg.svdvals(grad_to_2d(grads[i].cpu()))
+        ev_x = (s[:r_max] ** 2)
         rel = ((ev_r - ev_x).abs() / ev_x.clamp(min=1e-12)).max()
         assert rel < 1e-2, (spec.shape, float(rel))
