@@ -37,6 +37,7 @@ void atomo_batched_gram_launch(const float*, float*, const int64_t*,
 void atomo_batched_sel_launch(const float*, float*, const float*,
                               const int64_t*, const int32_t*, int, int,
                               hipStream_t);
+void atomo_jacobi_dense_launch(float*, float*, int, int, hipStream_t);
 void atomo_jacobi_eigh_launch(float*, float*, const int64_t*, const int64_t*,
                               const int32_t*, int, int, float*,
                               const int64_t*, int, hipStream_t);
@@ -288,6 +289,18 @@ void jacobi_eigh(torch::Tensor grams, torch::Tensor evals, torch::Tensor desc,
       (int)warm, cur_stream());
 }
 
+void jacobi_dense(torch::Tensor a, torch::Tensor evals, int64_t n_mats,
+                  int64_t nb) {
+  check_f32_cuda(a, "a");
+  check_f32_cuda(evals, "evals");
+  TORCH_CHECK(nb <= 64 && nb % 2 == 0, "nb must be even and <= 64");
+  TORCH_CHECK(a.numel() >= n_mats * nb * nb && evals.numel() >= n_mats * nb,
+              "jacobi_dense: buffer too small");
+  if (n_mats == 0) return;
+  atomo_jacobi_dense_launch(a.data_ptr<float>(), evals.data_ptr<float>(),
+                            (int)n_mats, (int)nb, cur_stream());
+}
+
 void jacobi_eigh_big(torch::Tensor grams, torch::Tensor vbuf,
                      torch::Tensor evals, torch::Tensor desc,
                      torch::Tensor eval_offs, torch::Tensor rows,
@@ -347,6 +360,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "batched parallel-Jacobi symmetric eigensolver (sm <= 64, LDS)");
   m.def("jacobi_eigh_big", &jacobi_eigh_big,
         "batched parallel-Jacobi eigensolver (64 < sm <= 512, L2-resident)");
+  m.def("jacobi_dense", &jacobi_dense,
+        "batched dense small (nb<=64) symmetric eigh, one wave per matrix; "
+        "eigenvalues descending, eigenvectors overwrite the input");
   m.def("build_stage", &build_stage,
         "gather sampled atoms into the staged wire factors");
   m.def("batched_gram", &batched_gram,

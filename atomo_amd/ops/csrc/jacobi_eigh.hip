@@ -214,6 +214,127 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// dense small-batch variant: a plain (n_mats, nb, nb) symmetric fp32 array,
+// one 64-thread wave per matrix, nb <= 64 even.  Eigenvalues sorted
+// DESCENDING into evals (n_mats, nb); eigenvectors overwrite the matrix
+// slot (column j = eigenvector j).  Serves the randomized big-fold solver
+// (svd_encoder._solve_big_folds_randomized): Lowdin orthonormalization and
+// the Rayleigh-Ritz projection both need tiny batched eighs, and batched
+// hipSOLVER syevd costs ~1 ms of launch latency per call — this kernel is
+// ~100 us for ~100 matrices.
+// ---------------------------------------------------------------------------
+#define DMAX 64
+#define DTHREADS 64
+
+__global__ void __launch_bounds__(DTHREADS) jacobi_dense_kernel(
+    float* __restrict__ a, float* __restrict__ evals, int n_mats, int nb) {
+  constexpr int DSTRIDE = DMAX + 1;
+  __shared__ float G[DMAX * DSTRIDE];
+  __shared__ float V[DMAX * DSTRIDE];
+  __shared__ float cs[DMAX / 2], sn[DMAX / 2];
+  __shared__ int pp[DMAX / 2], qq[DMAX / 2];
+  __shared__ int order[DMAX];
+  __shared__ float fro_s;
+  __shared__ int done_s;
+  if (blockIdx.x >= (unsigned)n_mats) return;
+  float* Ag = a + (int64_t)blockIdx.x * nb * nb;
+  float* ev = evals + (int64_t)blockIdx.x * nb;
+  const int tid = threadIdx.x;
+  const int N = nb;  // caller guarantees even, <= DMAX
+
+  float fro2 = 0.f;
+  for (int i = tid; i < N * N; i += DTHREADS) {
+    const int r = i / N, c = i % N;
+    const float g = Ag[i];
+    G[r * DSTRIDE + c] = g;
+    V[r * DSTRIDE + c] = (r == c) ? 1.f : 0.f;
+    fro2 += g * g;
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) fro2 += __shfl_xor(fro2, off, 64);
+  if (tid == 0) {
+    fro_s = fro2;
+    done_s = 0;
+  }
+  __syncthreads();
+  const float tol2 = fro_s * 1e-10f;
+
+  const int np = N / 2;
+  for (int sweep = 0; sweep < SWEEPS && !done_s; ++sweep) {
+    for (int round = 0; round < N - 1; ++round) {
+      if (tid < np) {
+        auto player = [&](int slot) {
+          return slot == 0 ? 0 : 1 + (slot - 1 + round) % (N - 1);
+        };
+        const int x = player(tid);
+        const int y = player(N - 1 - tid);
+        const int p = min(x, y), q = max(x, y);
+        pp[tid] = p;
+        qq[tid] = q;
+        rot_params(G[p * DSTRIDE + p], G[q * DSTRIDE + q], G[p * DSTRIDE + q],
+                   &cs[tid], &sn[tid]);
+      }
+      __syncthreads();
+      for (int i = tid; i < np * N; i += DTHREADS) {
+        const int pr = i / N, k = i % N;
+        const int p = pp[pr], q = qq[pr];
+        const float c = cs[pr], s = sn[pr];
+        const float gp = G[p * DSTRIDE + k], gq = G[q * DSTRIDE + k];
+        G[p * DSTRIDE + k] = c * gp - s * gq;
+        G[q * DSTRIDE + k] = s * gp + c * gq;
+      }
+      __syncthreads();
+      for (int i = tid; i < np * N; i += DTHREADS) {
+        const int pr = i / N, k = i % N;
+        const int p = pp[pr], q = qq[pr];
+        const float c = cs[pr], s = sn[pr];
+        const float gp = G[k * DSTRIDE + p], gq = G[k * DSTRIDE + q];
+        G[k * DSTRIDE + p] = c * gp - s * gq;
+        G[k * DSTRIDE + q] = s * gp + c * gq;
+        const float vp = V[k * DSTRIDE + p], vq = V[k * DSTRIDE + q];
+        V[k * DSTRIDE + p] = c * vp - s * vq;
+        V[k * DSTRIDE + q] = s * vp + c * vq;
+      }
+      __syncthreads();
+    }
+    float off2 = 0.f;
+    for (int i = tid; i < N * N; i += DTHREADS) {
+      const int r = i / N, c = i % N;
+      if (r != c) {
+        const float g = G[r * DSTRIDE + c];
+        off2 += g * g;
+      }
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) off2 += __shfl_xor(off2, off, 64);
+    if (tid == 0 && off2 <= tol2) done_s = 1;
+    __syncthreads();
+  }
+
+  if (tid == 0) {
+    for (int i = 0; i < N; ++i) order[i] = i;
+    for (int i = 1; i < N; ++i) {
+      const int oi = order[i];
+      const float vi = G[oi * DSTRIDE + oi];
+      int j = i - 1;
+      while (j >= 0 && G[order[j] * DSTRIDE + order[j]] < vi) {
+        order[j + 1] = order[j];
+        --j;
+      }
+      order[j + 1] = oi;
+    }
+  }
+  __syncthreads();
+  for (int j = tid; j < N; j += DTHREADS) ev[j] = G[order[j] * DSTRIDE + order[j]];
+  __syncthreads();
+  for (int i = tid; i < N * N; i += DTHREADS) {
+    const int k = i / N, j = i % N;
+    Ag[i] = V[k * DSTRIDE + order[j]];
+  }
+}
+
 // ---------------------------------------------------------------------------
 // big variant: G in its global Gram slot, V in global scratch
 // (both L2-resident for sm <= 512)
@@ -563,6 +684,12 @@ void atomo_jacobi_eigh_big_launch(float* grams, float* vbuf, float* evals,
   hipLaunchKernelGGL(jacobi_eigh_big_kernel, dim3(n_mats), dim3(JBIG_THREADS),
                      0, stream, grams, vbuf, evals, desc, eval_offs, rows_list,
                      v_offs, n_mats);
+}
+
+void atomo_jacobi_dense_launch(float* a, float* evals, int n_mats, int nb,
+                               hipStream_t stream) {
+  hipLaunchKernelGGL(jacobi_dense_kernel, dim3(n_mats), dim3(DTHREADS), 0,
+                     stream, a, evals, n_mats, nb);
 }
 
 void atomo_build_stage_launch(const float* evecs, const float* evals,

@@ -513,33 +513,86 @@ class BatchedSVDEncoder:
     # in _rsvd_tail_dev.  Atoms beyond the subspace read eval = 0 and are
     # never sampled.  (North-star "one-pass randomized SVD"; reference
     # semantics codings/svd.py:49-117.)
+    def _dense_eigh(self, s: torch.Tensor):
+        """Batched symmetric eigh via the one-wave-per-matrix LDS Jacobi
+        kernel (ops/csrc/jacobi_eigh.hip jacobi_dense_kernel).  ``s``
+        (N, b, b) contiguous fp32 cuda is OVERWRITTEN with eigenvectors
+        (column j = eigenvector j, descending); returns evals (N, b).
+        Replaces batched hipSOLVER syevd, whose ~1 ms per-call launch
+        latency dominated the randomized solve."""
+        from ..ops import ext
+
+        n, b = s.shape[0], s.shape[2]
+        ev = torch.empty(n, b, device=s.device)
+        ext().jacobi_dense(s, ev, n, b)
+        return ev, s
+
+    def _merged_lowdin(self, ys):
+        """Orthonormalize the columns of every (B_g, sm_g, b) panel:
+        Q = Y S^{-1/2} (Lowdin / symmetric orthogonalization) with S's eigh
+        merged across groups into ONE dense-Jacobi launch.  Column
+        normalization first so conditioning is angular, not scale."""
+        ys = [
+            y / y.norm(dim=1, keepdim=True).clamp(min=1e-30) for y in ys
+        ]
+        s = torch.cat(
+            [torch.bmm(y.transpose(1, 2), y) for y in ys], dim=0
+        ).contiguous()
+        lam, v = self._dense_eigh(s)
+        inv = lam.clamp(min=1e-6).rsqrt()
+        row0, out = 0, []
+        for y in ys:
+            B = y.shape[0]
+            vg = v[row0 : row0 + B]
+            m = torch.bmm(vg * inv[row0 : row0 + B].unsqueeze(1),
+                          vg.transpose(1, 2))
+            out.append(torch.bmm(y, m))
+            row0 += B
+        return out
+
     def _solve_big_folds_randomized(self) -> None:
         b = self._rsvd_b
-        group_ctx = []
-        ts = []
+        gs, qs, trs = [], [], []
+        cold = False
         for sm, idxs, gather, scatter, evi, tails in self._rsvd_groups:
             B = len(idxs)
             g = self.grams.index_select(0, gather).view(B, sm, sm)
             g = 0.5 * (g + g.transpose(1, 2))
-            tr = g.diagonal(dim1=1, dim2=2).sum(dim=1)
+            trs.append(g.diagonal(dim1=1, dim2=2).sum(dim=1))
             q = self._rsvd_Q.get(sm)
-            iters = 1
             if q is None or q.shape[0] != B:
                 q = torch.randn(
                     B, sm, b, generator=self._rsvd_gen, device=self.device
                 )
-                iters = 2  # cold start: extra power step for subspace quality
-            t, q = subspace_iterate(g, q, iters)
-            ts.append(t)
-            group_ctx.append((sm, q, tr, scatter, evi, tails))
-        evals_t, w = _robust_eigh(torch.cat(ts, dim=0), out_dtype=torch.float32)
-        evals_t = evals_t.flip(1).clamp(min=0.0)
-        w = w.flip(2)
+                cold = True
+            gs.append(g)
+            qs.append(q)
+        # warm-started subspace iteration: power step(s) + Lowdin, then a
+        # second Lowdin pass for near-machine orthogonality (the decoded
+        # atom is A v v^T / p: unbiased for the projection onto span(V)
+        # only when V is orthonormal)
+        for _ in range(2 if cold else 1):
+            qs = self._merged_lowdin(
+                [torch.bmm(g, q) for g, q in zip(gs, qs)]
+            )
+        qs = self._merged_lowdin(qs)
+        # Rayleigh-Ritz: T = Q^T G Q, merged eigh, Ritz pairs (lam, Q W)
+        ts = torch.cat(
+            [
+                torch.bmm(q.transpose(1, 2), torch.bmm(g, q))
+                for g, q in zip(gs, qs)
+            ],
+            dim=0,
+        ).contiguous()
+        lam_all, w_all = self._dense_eigh(ts)
+        lam_all = lam_all.clamp(min=0.0)
         row0 = 0
-        for sm, q, tr, scatter, evi, tails in group_ctx:
+        for (sm, idxs, gather, scatter, evi, tails), q, tr in zip(
+            self._rsvd_groups, qs, trs
+        ):
             B = q.shape[0]
-            lam = evals_t[row0 : row0 + B]  # (B, b) descending
-            evecs = torch.bmm(q, w[row0 : row0 + B])  # (B, sm, b) orthonormal
+            lam = lam_all[row0 : row0 + B]  # (B, b) descending
+            evecs = torch.bmm(q, w_all[row0 : row0 + B])
             row0 += B
             self._rsvd_Q[sm] = evecs  # warm subspace for the next step
             # single-kernel writebacks (evals tail slots stay zero from
