@@ -313,16 +313,20 @@ class BatchedSVDEncoder:
                 self.vwarm = torch.zeros(max(1, vw), dtype=torch.float32, device=dev)
                 self.vwarm_offs = torch.tensor(vw_offs, dtype=torch.int64, device=dev)
                 self._warm = False
-                # selective-overlap support: backward hooks cover only the
-                # big-fold (rocBLAS mm) and host layers; j64 grams come from
-                # ONE batched_gram launch post-backward — hundreds of
-                # per-layer Python hooks during backward cost more host time
-                # than the single kernel (ResNet-152: 467 hooks -> 97).
+                # selective-overlap support ("big" hook mode): backward
+                # hooks cover only the layers whose post-gram work benefits
+                # from launching early (host-LAPACK layers; plus the big
+                # folds in exact-syevd mode, whose solver groups fire from
+                # hooks).  Everything else runs post-backward in a handful
+                # of batched launches — hundreds of per-layer Python hooks
+                # cost ~40 us each of host time during backward.
                 # batched_gram accumulates atomically, so the j64 slots are
                 # zeroed selectively via this index when hooks own the rest.
-                self.hook_layers = set(self.big_gram_layers) | {
+                self.hook_layers = {
                     i for i in range(len(specs)) if i not in self.kernel_set
                 }
+                if self.exact_eigh:
+                    self.hook_layers |= set(self.big_gram_layers)
                 self.hook_mode = "all"  # set by setup_overlap
                 j64_ids = [row_to_layer[r] for r in rows_j64]
                 self._j64_zero_idx = (
@@ -338,6 +342,33 @@ class BatchedSVDEncoder:
                     if j64_ids
                     else None
                 )
+                # big-fold Grams batched by identical fold shape: ONE
+                # stack + bmm + scatter per shape group (ResNet-152 has 97
+                # big folds in ~8 shapes) instead of a per-layer mm loop
+                self._gram_shape_groups = []
+                if not self.exact_eigh and self.big_gram_layers:
+                    by_shape = defaultdict(list)
+                    for i in self.big_gram_layers:
+                        s = specs[i]
+                        by_shape[(s.meta["m"], s.meta["n"])].append(i)
+                    for (m, n), idxs in sorted(by_shape.items()):
+                        sm = min(m, n)
+                        scat = torch.cat(
+                            [
+                                torch.arange(sm * sm, dtype=torch.int64)
+                                + self.gram_offsets[i]
+                                for i in idxs
+                            ]
+                        ).to(dev)
+                        self._gram_shape_groups.append(
+                            (
+                                m,
+                                n,
+                                m >= n,
+                                [param_offsets[i] for i in idxs],
+                                scat,
+                            )
+                        )
                 self.eval_offs_dev = torch.tensor(eval_offs, dtype=torch.int64, device=dev)
                 self.eval_offs = eval_offs
                 self.evals_dev = torch.zeros(max(1, ev_off), dtype=torch.float32, device=dev)
@@ -513,6 +544,19 @@ class BatchedSVDEncoder:
     # in _rsvd_tail_dev.  Atoms beyond the subspace read eval = 0 and are
     # never sampled.  (North-star "one-pass randomized SVD"; reference
     # semantics codings/svd.py:49-117.)
+    def _compute_big_grams(self, flat_grad: torch.Tensor) -> None:
+        """Batched big-fold Grams: one stack + bmm + scatter per identical
+        fold shape (randomized mode; replaces the per-layer rocBLAS loop)."""
+        for m, n, tall, offs, scat in self._gram_shape_groups:
+            a = torch.stack(
+                [flat_grad.narrow(0, o, m * n) for o in offs]
+            ).view(len(offs), m, n)
+            if tall:
+                g = torch.bmm(a.transpose(1, 2), a)
+            else:
+                g = torch.bmm(a, a.transpose(1, 2))
+            self.grams.index_copy_(0, scat, g.reshape(-1))
+
     def _dense_eigh(self, s: torch.Tensor):
         """Batched symmetric eigh via the one-wave-per-matrix LDS Jacobi
         kernel (ops/csrc/jacobi_eigh.hip jacobi_dense_kernel).  ``s``
@@ -789,7 +833,13 @@ class BatchedSVDEncoder:
             ov_host = self._ov_host_results
         host_layers = [i for i in range(len(specs)) if i not in kernel_set]
         host_unsolved = [i for i in host_layers if i not in ov_host]
-        mm_layers = host_layers + (self.big_gram_layers if use_kernels else [])
+        # big-fold grams: per-layer mm only in exact mode (randomized mode
+        # batches them by shape in _compute_big_grams)
+        mm_layers = host_layers + (
+            self.big_gram_layers
+            if use_kernels and self.exact_eigh
+            else []
+        )
         a2ds = {i: self._a2d(grads[i], specs[i]) for i in mm_layers}
 
         # ---- phase A: Grams (batched kernel + rocBLAS leftovers) -------
@@ -804,9 +854,12 @@ class BatchedSVDEncoder:
                         flat_grad, self.grams, self.desc, self.gram_work,
                         self.gram_work.shape[0],
                     )
+                self._compute_big_grams(flat_grad)
             elif self.hook_mode == "big":
-                # hooks covered only big-fold/host grams; j64 grams run here
-                # as one launch (selective zero first: atomic accumulation)
+                # hooks covered only their layers (host / exact-mode big
+                # folds); j64 grams run here as one launch (selective zero
+                # first: atomic accumulation) and randomized-mode big folds
+                # as shape-batched bmms
                 if self._j64_zero_idx is not None:
                     self.grams.index_fill_(0, self._j64_zero_idx, 0.0)
                 if self.gram_work.shape[0]:
@@ -814,6 +867,7 @@ class BatchedSVDEncoder:
                         flat_grad, self.grams, self.desc, self.gram_work,
                         self.gram_work.shape[0],
                     )
+                self._compute_big_grams(flat_grad)
         for i in (() if grams_done else mm_layers):
             a = a2ds[i]
             sm = self.small[i]
