@@ -17,6 +17,8 @@ from __future__ import annotations
 
 import datetime
 import os
+import time
+from collections import deque
 from typing import Optional
 
 import torch
@@ -129,6 +131,222 @@ class Comm:
         dist.isend(send, dst=dst).wait()
         return None
 
+    # -- true partial aggregation ----------------------------------------
+    # --num-aggregate K must RETURN after K
+    # contributions so a straggling rank stops costing wall-clock (the
+    # reference stores the flag but never implements it,
+    # sync_replicas_master_nn.py:113,124).  Design: a depth-D pipeline
+    # (ATOMO_PIPELINE_DEPTH, default 6) with STRICT 1:1 per-step message
+    # matching — every rank posts exactly one send and the PS one receive
+    # per (worker, step), so ordering assigns each packet to its step and
+    # draining can never deadlock.  The PS decodes current-step arrivals
+    # as they land, returns after K, and leaves the rest pending; a
+    # packet observed late is dropped and counted in ``stale_drops``.  A
+    # straggler up to D-1 steps behind costs the PS nothing; beyond that
+    # the PS throttles to the straggler's pace (tolerating unbounded lag
+    # needs step-skipping semantics the synchronous contract forbids).
+    # In this mode the weight push is a per-worker isend pipeline too
+    # (ps_push_weights / recv_weights) — a broadcast collective would
+    # rendezvous with the sleeping rank every step and stall the PS.
+    PIPE_DEPTH = int(os.environ.get("ATOMO_PIPELINE_DEPTH", "6"))
+
+    def gather_partial(
+        self,
+        send: torch.Tensor,
+        bufs: Optional[torch.Tensor],  # (D, world, words) on dst
+        step: int,
+        dst: int = 0,
+        target: int = 0,
+        on_arrival=None,  # called as on_arrival(worker, slot)
+        self_counts: bool = True,
+    ) -> int:
+        """Arrival-order gather that returns after ``target`` current-step
+        contributions (0 = all).  Returns the contribution count on dst,
+        0 elsewhere."""
+        if not self._initialized:
+            if bufs is not None:
+                bufs[0, 0].copy_(send)
+                if on_arrival is not None:
+                    on_arrival(0, 0)
+            return 1
+        d = self.PIPE_DEPTH
+        sl = step % d
+        if self.rank == dst:
+            # completion signaling differs per backend: NCCL/RCCL works
+            # carry device events, so is_completed() polling is reliable;
+            # gloo works only resolve inside wait() (is_completed() stays
+            # False forever in this torch build), so a per-worker waiter
+            # thread blocks in wait() and posts (worker, step) arrival
+            # events to a queue the PS consumes.
+            threaded = self.backend != "nccl"
+            if not hasattr(self, "_pg_recv"):
+                import queue as _queue
+                import threading as _threading
+
+                self._pg_recv = {}  # w -> deque of (req, step)  [poll mode]
+                self.stale_drops = 0
+                self._pg_landed = {}  # w -> newest landed step [thread mode]
+                self._pg_outstanding = 0  # posted - landed [thread mode]
+                if threaded:
+                    self._ev_q = _queue.Queue()
+                    self._wt_q = {}
+                    self._wt = []
+                    for w in range(self.world):
+                        if w == dst:
+                            continue
+                        wq = _queue.Queue()
+                        self._wt_q[w] = wq
+
+                        def waiter(wq=wq, evq=self._ev_q):
+                            while True:
+                                item = wq.get()
+                                if item is None:
+                                    return
+                                req, rstep, ww = item
+                                req.wait()
+                                evq.put((ww, rstep))
+
+                        th = _threading.Thread(target=waiter, daemon=True)
+                        th.start()
+                        self._wt.append(th)
+            workers = [w for w in range(self.world) if w != dst]
+            contrib = 0
+            want = target if target > 0 else (
+                len(workers) + (1 if self_counts else 0)
+            )
+
+            def consume(w, rstep):
+                nonlocal contrib
+                if threaded:
+                    self._pg_outstanding -= 1
+                self._pg_landed[w] = rstep
+                if rstep == step:
+                    contrib += 1
+                    if on_arrival is not None:
+                        on_arrival(w, sl)
+                else:
+                    self.stale_drops += 1
+
+            # recycle this slot: the receive posted D steps ago must have
+            # LANDED before its buffer is reposted (bounded staleness D)
+            if threaded:
+                if not hasattr(self, "_pg_first"):
+                    self._pg_first = {}
+                for w in workers:
+                    first = self._pg_first.setdefault(w, step)
+                    # per-worker receives land in order, so landed[w]
+                    # >= step-d means the slot's old receive is done
+                    while (
+                        step - d >= first
+                        and self._pg_landed.get(w, first - 1) < step - d
+                    ):
+                        ww, rstep = self._ev_q.get()
+                        consume(ww, rstep)
+                for w in workers:
+                    req = dist.irecv(bufs[sl, w], src=w)
+                    self._wt_q[w].put((req, step, w))
+                    self._pg_outstanding += 1
+            else:
+                for w in workers:
+                    dq = self._pg_recv.setdefault(w, deque())
+                    while dq and dq[0][1] <= step - d:
+                        rstep = dq[0][1]
+                        dq.popleft()[0].wait()
+                        consume(w, rstep)
+                    dq.append((dist.irecv(bufs[sl, w], src=w), step))
+            if self_counts:
+                bufs[sl, dst].copy_(send)
+                contrib += 1
+                if on_arrival is not None:
+                    on_arrival(dst, sl)
+            if threaded:
+                while contrib < want:
+                    ww, rstep = self._ev_q.get()
+                    consume(ww, rstep)
+            else:
+                while contrib < want:
+                    progressed = False
+                    for w in workers:
+                        dq = self._pg_recv[w]
+                        # drain in posted order; per-pair ordering means
+                        # the head completes first
+                        while dq and dq[0][0].is_completed():
+                            _, rstep = dq.popleft()
+                            progressed = True
+                            consume(w, rstep)
+                        if contrib >= want:
+                            break
+                    if not progressed and contrib < want:
+                        time.sleep(0)
+            return contrib
+        # worker: pipelined isend (the wire is rewritten next step while
+        # this send may still be in flight)
+        if not hasattr(self, "_pg_send"):
+            self._pg_send = deque()
+        if len(self._pg_send) >= d:
+            req, buf = self._pg_send.popleft()
+            req.wait()
+        else:
+            buf = torch.empty_like(send)
+        buf.copy_(send)
+        self._pg_send.append((dist.isend(buf, dst=dst), buf))
+        return 0
+
+    def ps_push_weights(self, flat: torch.Tensor, step: int) -> None:
+        """PS-side per-worker weight push for partial mode: one pipelined
+        isend per worker per step (1:1 with recv_weights calls); the send
+        buffer for a slot is recycled once its D-steps-ago sends finish."""
+        if not self._initialized:
+            return
+        d = self.PIPE_DEPTH
+        if not hasattr(self, "_wpush"):
+            self._wpush = [None] * d  # slot -> (reqs, buf)
+        sl = step % d
+        ent = self._wpush[sl]
+        if ent is not None:
+            for r in ent[0]:
+                r.wait()
+            buf = ent[1]
+        else:
+            buf = torch.empty_like(flat)
+        buf.copy_(flat)
+        reqs = [
+            dist.isend(buf, dst=w) for w in range(self.world) if w != self.rank
+        ]
+        self._wpush[sl] = (reqs, buf)
+
+    def recv_weights(self, flat: torch.Tensor, src: int = 0) -> None:
+        """Worker-side blocking weight receive (matches ps_push_weights
+        1:1 per step; a late worker just consumes its queued packets in
+        order at its own pace)."""
+        if self._initialized:
+            dist.recv(flat, src=src)
+
+    def drain_partial(self) -> None:
+        """Complete everything left pending by the partial-mode pipelines
+        (safe by 1:1 matching; call before destroying the group)."""
+        for dq in getattr(self, "_pg_recv", {}).values():
+            while dq:
+                dq.popleft()[0].wait()
+        n = getattr(self, "_pg_outstanding", 0)
+        while n > 0:
+            self._ev_q.get()
+            n -= 1
+        self._pg_outstanding = 0
+        for wq in getattr(self, "_wt_q", {}).values():
+            wq.put(None)  # stop waiter threads
+        self._wt_q = {}
+        if hasattr(self, "_pg_recv"):
+            del self._pg_recv  # force re-init on next partial use
+        for ent in getattr(self, "_pg_send", []):
+            ent[0].wait()
+        self._pg_send = deque()
+        for ent in getattr(self, "_wpush", []):
+            if ent is not None:
+                for r in ent[0]:
+                    r.wait()
+        self._wpush = [None] * self.PIPE_DEPTH
+
     def barrier(self) -> None:
         if self._initialized:
             if self.backend == "nccl":
@@ -138,5 +356,6 @@ class Comm:
 
     def close(self) -> None:
         if self._initialized and dist.is_initialized():
+            self.drain_partial()
             dist.destroy_process_group()
             self._initialized = False

@@ -129,6 +129,20 @@ class PSTrainer:
         )
         if self.p2p and self.wc.reducible:
             self.p2p = False  # raw codec always rides the RCCL reduce
+        # depth-D pipelined receive buffers for TRUE partial aggregation
+        # (the PS returns after K arrivals; late receives stay pending
+        # across steps and are dropped as stale when they land)
+        self.pp_buf = (
+            torch.zeros(
+                comm.PIPE_DEPTH,
+                comm.world,
+                self.wc.total_words,
+                dtype=torch.float32,
+                device=self.device,
+            )
+            if self.p2p and num_aggregate and self.is_master
+            else None
+        )
         self.agg = (
             torch.zeros_like(self.flat)
             if self.is_master and not self.wc.reducible
@@ -195,7 +209,16 @@ class PSTrainer:
     def train_step(self, x: torch.Tensor, y: torch.Tensor) -> float:
         t = self.timers
         with t.phase("fetch"):
-            self.comm.broadcast(self.flat, src=0)
+            if self.p2p and self.num_aggregate > 0:
+                # partial mode: per-worker pipelined weight push — a
+                # broadcast collective would rendezvous with a straggler
+                # every step and stall the PS (see Comm.gather_partial)
+                if self.is_master:
+                    self.comm.ps_push_weights(self.flat, self.step_num)
+                else:
+                    self.comm.recv_weights(self.flat, src=0)
+            else:
+                self.comm.broadcast(self.flat, src=0)
 
         if self.is_worker:
             with t.phase("comp"):
@@ -235,28 +258,59 @@ class PSTrainer:
             self.flat_grad.zero_()  # dedicated PS contributes zeros to the sum
 
         if self.p2p and not self.wc.reducible:
-            # arrival-order gather: the PS decodes each worker's bucket as
-            # it lands, overlapping decode with the remaining receives
             with t.phase("comm"):
-                contrib = [0]
-                target = self.num_aggregate or self.num_workers
-                if self.is_master:
-                    self.agg.zero_()
+                if self.num_aggregate > 0:
+                    # TRUE partial aggregation: the PS returns after the
+                    # first K contributions; a straggler's packet is left
+                    # pending and dropped (stale) when it lands next step
+                    contrib = [0]
+                    if self.is_master:
+                        self.agg.zero_()
 
-                    def on_arrival(w):
-                        if self.dedicated_ps and w == 0:
-                            return
-                        if contrib[0] < target:
+                        def on_arr(w, ph):
+                            self.wc.decode_all(
+                                self.pp_buf[ph, w : w + 1], self.agg
+                            )
+                            contrib[0] += 1
+
+                        contrib[0] = 0
+                        self.comm.gather_partial(
+                            self.wire,
+                            self.pp_buf,
+                            self.step_num,
+                            dst=0,
+                            target=self.num_aggregate,
+                            on_arrival=on_arr,
+                            self_counts=not self.dedicated_ps,
+                        )
+                    else:
+                        self.comm.gather_partial(
+                            self.wire, None, self.step_num, dst=0,
+                            target=self.num_aggregate,
+                        )
+                        contrib[0] = self.num_aggregate
+                else:
+                    # full-sync arrival-order gather: the PS decodes each
+                    # worker's bucket as it lands, overlapping decode with
+                    # the remaining receives
+                    contrib = [0]
+                    if self.is_master:
+                        self.agg.zero_()
+
+                        def on_arrival(w):
+                            if self.dedicated_ps and w == 0:
+                                return
                             self.wc.decode_all(
                                 self.gather_buf[w : w + 1], self.agg
                             )
                             contrib[0] += 1
 
-                    self.comm.gather_arrival(
-                        self.wire, self.gather_buf, dst=0, on_arrival=on_arrival
-                    )
-                else:
-                    self.comm.gather_arrival(self.wire, None, dst=0)
+                        self.comm.gather_arrival(
+                            self.wire, self.gather_buf, dst=0,
+                            on_arrival=on_arrival,
+                        )
+                    else:
+                        self.comm.gather_arrival(self.wire, None, dst=0)
             self._last_contrib = contrib[0]
             grad_flat = self.agg if self.is_master else None
         else:

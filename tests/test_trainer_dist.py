@@ -153,3 +153,68 @@ def test_dist_p2p_mode(num_agg, port):
     # contribution count: all 3 (colocated) or capped at num_aggregate
     expect = 3 if num_agg == 0 else num_agg
     assert results[0][2] == expect
+
+
+def _run_rank_straggler(rank, world, port, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.set_num_threads(2)
+    import time
+
+    from atomo_amd.codings import make_codec
+    from atomo_amd.data import make_loaders
+    from atomo_amd.parallel import Comm, PSTrainer
+
+    comm = Comm(backend="gloo", device=torch.device("cpu"))
+    trainer = PSTrainer(
+        model_name="LeNet",
+        codec=make_codec("svd", rank=3),
+        comm=comm, lr=0.05, momentum=0.9, num_classes=10, in_channels=1,
+        seed=7, device=torch.device("cpu"), comm_type="P2P", num_aggregate=2,
+    )
+    train, _ = make_loaders("mnist", 16, 16, torch.device("cpu"), seed=50 + rank)
+    it = iter(train)
+    durations = []
+    for i in range(8):
+        x, y = next(it)
+        if rank == 2 and 2 <= i <= 5:
+            time.sleep(0.4)  # persistent straggler for 4 steps
+        t0 = time.perf_counter()
+        trainer.train_step(x, y)
+        durations.append(time.perf_counter() - t0)
+    stale = getattr(comm, "stale_drops", 0)
+    q.put((rank, durations, stale, trainer.flat.sum().item()))
+    comm.barrier()
+    comm.close()
+
+
+def test_dist_partial_aggregation_drops_straggler():
+    """with --num-aggregate 2 and 3 ranks, a rank that
+    sleeps 400 ms per step must NOT add 400 ms to the PS's step time —
+    the PS returns after the 2 fastest contributions and the late packet
+    is dropped as stale."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    world, port = 3, 29631
+    procs = [
+        ctx.Process(target=_run_rank_straggler, args=(r, world, port, q))
+        for r in range(world)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, durations, stale, fsum = q.get(timeout=240)
+        results[rank] = (durations, stale, fsum)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    ps_durs = results[0][0]
+    # straggler steps on the PS: well under the 400 ms sleep (full-sync
+    # would absorb ~400 ms each).  Bounded staleness (2) means at most the
+    # FIRST straggler step can stall the pipeline; steps 3.. must be fast.
+    slow_window = ps_durs[3:6]
+    assert max(slow_window) < 0.3, ps_durs
+    assert results[0][1] > 0  # stale packets were observed and dropped
