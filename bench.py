@@ -98,6 +98,59 @@ def main():
         if device.type == "cuda":
             torch.cuda.synchronize()
 
+    if comm.world > 1:
+        # N>=2 pre-flight: verify every collective the step loop uses
+        # (broadcast / gather / reduce) round-trips correctly under this
+        # backend BEFORE the timed region, so an 8-GPU run fails loudly
+        # in seconds rather than producing garbage
+        import sys
+
+        probe = torch.full((1024,), float(comm.rank + 1), device=device)
+        bc = probe.clone()
+        comm.broadcast(bc, src=0)
+        ok_bc = bool((bc == 1.0).all())
+        stack = (
+            torch.zeros(comm.world, 1024, device=device)
+            if comm.rank == 0
+            else None
+        )
+        comm.gather(probe, stack, dst=0)
+        ok_g = comm.rank != 0 or all(
+            bool((stack[w] == float(w + 1)).all()) for w in range(comm.world)
+        )
+        red = probe.clone()
+        comm.reduce_sum(red, dst=0)
+        expect = sum(range(1, comm.world + 1))
+        ok_r = comm.rank != 0 or bool((red == float(expect)).all())
+        if not (ok_bc and ok_g and ok_r):
+            print(
+                json.dumps(
+                    {
+                        "preflight": "FAIL",
+                        "rank": comm.rank,
+                        "broadcast": ok_bc,
+                        "gather": ok_g,
+                        "reduce": ok_r,
+                        "backend": comm.backend,
+                    }
+                ),
+                file=sys.stderr,
+                flush=True,
+            )
+            raise SystemExit(2)
+        if comm.rank == 0:
+            print(
+                json.dumps(
+                    {
+                        "preflight": "ok",
+                        "world": comm.world,
+                        "backend": comm.backend,
+                    }
+                ),
+                file=sys.stderr,
+                flush=True,
+            )
+
     it = iter_cycle(train)
     for _ in range(a.warmup):
         x, y = next(it)
@@ -114,11 +167,17 @@ def main():
         trainer.train_step(x, y)
     sync()
     elapsed = time.perf_counter() - t0
-    if a.phase_log:
+    if a.phase_log or comm.world > 1:
+        # per-rank phase breakdown (always on for N>1: the comm phase per
+        # rank is the scaling diagnostic).  Without --phase-log these are
+        # host-side enqueue times for async device work.
         import sys
 
-        print(json.dumps({"rank": comm.rank, **trainer.timers.summary()}),
-              file=sys.stderr, flush=True)
+        rec = {"rank": comm.rank, **trainer.timers.summary()}
+        sd = getattr(comm, "stale_drops", 0)
+        if sd:
+            rec["stale_drops"] = sd
+        print(json.dumps(rec), file=sys.stderr, flush=True)
 
     # max over ranks
     t = torch.tensor([elapsed], dtype=torch.float64,
