@@ -49,7 +49,8 @@ void atomo_build_stage_launch(const float*, const float*, const float*,
                               hipStream_t);
 void atomo_sample_stage_launch(const float*, const float*, float*,
                                const int64_t*, const int64_t*, int, int, int,
-                               uint64_t, unsigned long long*, hipStream_t);
+                               const unsigned long long*,
+                               unsigned long long*, hipStream_t);
 }
 
 namespace {
@@ -334,10 +335,13 @@ void build_stage(torch::Tensor evecs, torch::Tensor evals,
 void sample_stage(torch::Tensor evecs, torch::Tensor evals,
                   torch::Tensor stage, torch::Tensor desc,
                   torch::Tensor eval_offs, int64_t n_layers, int64_t rank,
-                  bool truncate, int64_t seed, torch::Tensor used_words) {
+                  bool truncate, torch::Tensor seed_dev,
+                  torch::Tensor used_words) {
   check_f32_cuda(evecs, "evecs");
   check_f32_cuda(evals, "evals");
   check_f32_cuda(stage, "stage");
+  TORCH_CHECK(seed_dev.is_cuda() && seed_dev.scalar_type() == torch::kInt64,
+              "seed_dev must be cuda int64");
   TORCH_CHECK(used_words.is_cuda() &&
                   used_words.scalar_type() == torch::kInt64,
               "used_words must be cuda int64");
@@ -346,7 +350,9 @@ void sample_stage(torch::Tensor evecs, torch::Tensor evals,
       evecs.data_ptr<float>(), evals.data_ptr<float>(),
       stage.data_ptr<float>(), desc.data_ptr<int64_t>(),
       eval_offs.data_ptr<int64_t>(), (int)n_layers, (int)rank,
-      truncate ? 1 : 0, (uint64_t)seed,
+      truncate ? 1 : 0,
+      reinterpret_cast<const unsigned long long*>(
+          seed_dev.data_ptr<int64_t>()),
       reinterpret_cast<unsigned long long*>(used_words.data_ptr<int64_t>()),
       cur_stream());
 }

@@ -539,8 +539,12 @@ __global__ void __launch_bounds__(64) sample_stage_kernel(
     const float* __restrict__ evecs, const float* __restrict__ evals,
     float* __restrict__ stage, const int64_t* __restrict__ desc,
     const int64_t* __restrict__ eval_offs, int n_layers, int rank,
-    int truncate, uint64_t seed,
+    int truncate, const unsigned long long* __restrict__ seed_buf,
     unsigned long long* __restrict__ used_words) {
+  // seed read from device memory so a hipGraph replay draws fresh dice:
+  // the host advances a pinned scalar, a captured H2D copy refreshes
+  // seed_buf, and every replay samples new atoms
+  const uint64_t seed = seed_buf[0];
   __shared__ float s_lds[SAMPLE_SM_MAX];
   __shared__ float s_sel[R_CAP], inv_s[R_CAP];
   __shared__ int idxs[R_CAP];
@@ -641,12 +645,13 @@ extern "C" {
 void atomo_sample_stage_launch(const float* evecs, const float* evals,
                                float* stage, const int64_t* desc,
                                const int64_t* eval_offs, int n_layers,
-                               int rank, int truncate, uint64_t seed,
+                               int rank, int truncate,
+                               const unsigned long long* seed_buf,
                                unsigned long long* used_words,
                                hipStream_t stream) {
   hipLaunchKernelGGL(sample_stage_kernel, dim3(n_layers), dim3(64), 0, stream,
                      evecs, evals, stage, desc, eval_offs, n_layers, rank,
-                     truncate, seed, used_words);
+                     truncate, seed_buf, used_words);
 }
 
 

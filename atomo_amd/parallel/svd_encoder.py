@@ -379,6 +379,12 @@ class BatchedSVDEncoder:
                 self.grams_host = torch.zeros(gram_off, dtype=torch.float32, pin_memory=True)
                 self.sel_elems = sel_elems
                 self.used_words_dev = torch.zeros(1, dtype=torch.int64, device=dev)
+                # sampler seed rides device memory (refreshed by a
+                # captured H2D copy from pinned host) so hipGraph replays
+                # draw fresh atoms; advance_seed() is the host-side LCG
+                self.seed_host = torch.zeros(1, dtype=torch.int64,
+                                             pin_memory=True)
+                self.seed_dev = torch.zeros(1, dtype=torch.int64, device=dev)
                 # rank-mixed: each worker's on-device Bernoulli sampler draws
                 # independent atoms (identical seeds collapse the
                 # 1/W compression-variance reduction of PS averaging)
@@ -544,6 +550,16 @@ class BatchedSVDEncoder:
     # in _rsvd_tail_dev.  Atoms beyond the subspace read eval = 0 and are
     # never sampled.  (North-star "one-pass randomized SVD"; reference
     # semantics codings/svd.py:49-117.)
+    def advance_seed(self) -> None:
+        """Host-side LCG step for the on-device sampler seed.  Under a
+        whole-step hipGraph the trainer calls this before each replay:
+        the captured seed_dev<-seed_host copy re-reads the pinned scalar
+        at replay time, so every replay samples new atoms."""
+        self._seed = (
+            self._seed * 6364136223846793005 + 1442695040888963407
+        ) % (1 << 62)
+        self.seed_host[0] = self._seed
+
     def _compute_big_grams(self, flat_grad: torch.Tensor) -> None:
         """Batched big-fold Grams: one stack + bmm + scatter per identical
         fold shape (randomized mode; replaces the per-layer rocBLAS loop)."""
@@ -935,13 +951,13 @@ class BatchedSVDEncoder:
         device_sampled = use_kernels and self.codec.generator is None
         self.device_counted = device_sampled
         if device_sampled:
-            self._seed = (self._seed * 6364136223846793005 + 1442695040888963407) % (
-                1 << 62
-            )
+            if not torch.cuda.is_current_stream_capturing():
+                self.advance_seed()
+            self.seed_dev.copy_(self.seed_host, non_blocking=True)
             e.sample_stage(
                 self.grams, self.evals_dev, self.stage_dev, self.desc,
                 self.eval_offs_dev, len(self.kernel_rows),
-                self.codec.rank, not self.codec.random_sample, self._seed,
+                self.codec.rank, not self.codec.random_sample, self.seed_dev,
                 self.used_words_dev,
             )
             e.batched_sel(

@@ -187,6 +187,25 @@ class PSTrainer:
         self._static_x = None
         self._static_y = None
         self._static_loss = None
+        # WHOLE-STEP graph (fwd/bwd + encode + decode + apply in ONE
+        # replay): eligible when the step is a pure device-kernel pipeline
+        # — single process, raw codec (wire aliases grads) or the fully-
+        # device SVD path (device sampler seed rides pinned memory so each
+        # replay draws fresh atoms).  Collectives are no-ops at world==1.
+        self._wgraph = None
+        self._wgraph_lr = None
+        self.graph_whole = False
+        if self.use_graph and comm.world == 1 and not self.dedicated_ps:
+            if self.wc.reducible:
+                self.graph_whole = True
+            else:
+                enc = self.wc._batched_encoder
+                self.graph_whole = bool(
+                    enc is not None
+                    and enc.use_kernels
+                    and len(enc.kernel_set) == len(self.wc.specs)
+                    and self.codec.generator is None
+                )
         # backward-hook overlap (reference *Split capability): per-layer
         # encode work on a side stream while backward continues.
         self.overlap = (
@@ -208,6 +227,24 @@ class PSTrainer:
     # -----------------------------------------------------------------
     def train_step(self, x: torch.Tensor, y: torch.Tensor) -> float:
         t = self.timers
+        if self.graph_whole:
+            with t.phase("comp"):
+                done = self._whole_step_graphed(x, y)
+            if done:
+                if self.wc.reducible:
+                    t.add_scalar("msg_bytes", 4.0 * self.wc.total_words)
+                self._last_contrib = self.num_workers
+                self.step_num += 1
+                if self.watchdog is not None:
+                    self.watchdog.step()
+                if self.step_num % self.shrink_freq == 0:
+                    self.lr *= self.lr_shrinkage
+                if (
+                    self.checkpoint_freq > 0
+                    and self.step_num % self.checkpoint_freq == 0
+                ):
+                    self.save_checkpoint()
+                return self.last_loss
         with t.phase("fetch"):
             if self.p2p and self.num_aggregate > 0:
                 # partial mode: per-worker pipelined weight push — a
@@ -357,6 +394,63 @@ class PSTrainer:
         if self._loss_tensor is not None:
             self.last_loss = float(self._loss_tensor)
         return self.last_loss
+
+    def _step_body(self) -> None:
+        """One full training step on static buffers (the captured body)."""
+        self.flat_grad.zero_()
+        loss = self.loss_fn(self.model(self._static_x), self._static_y)
+        loss.backward()
+        self._static_loss = loss.detach()
+        if self.wc.reducible:
+            grad = self.flat_grad
+        else:
+            self.wc.encode_all(self.wire, flat_grad=self.flat_grad)
+            self.agg.zero_()
+            self.wc.decode_all(self.wire.view(1, -1), self.agg)
+            grad = self.agg
+        self._apply(grad)
+
+    def _whole_step_graphed(self, x: torch.Tensor, y: torch.Tensor) -> bool:
+        """Capture-once/replay whole-step graph.  Returns False (and
+        disables itself) if capture fails — caller falls back to the
+        standard path."""
+        enc = self.wc._batched_encoder
+        if self._wgraph is None or self._wgraph_lr != self.lr:
+            try:
+                self._wgraph = None
+                self.opt.lr = self.lr
+                self._static_x = x.clone()
+                self._static_y = y.clone()
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    for _ in range(3):  # MIOpen algo search + warm states
+                        self._step_body()
+                torch.cuda.current_stream().wait_stream(side)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._step_body()
+                self._wgraph = g
+                self._wgraph_lr = self.lr
+            except Exception as exc:
+                print(
+                    f"[atomo] whole-step hipGraph capture failed ({exc}); "
+                    "standard path",
+                    flush=True,
+                )
+                self.graph_whole = False
+                self._wgraph = None
+                return False
+        self._static_x.copy_(x)
+        self._static_y.copy_(y)
+        if enc is not None:
+            enc.advance_seed()  # captured H2D copy re-reads the pinned seed
+        self._wgraph.replay()
+        if self.defer_loss:
+            self._loss_tensor = self._static_loss
+        else:
+            self.last_loss = float(self._static_loss)
+        return True
 
     def _fwd_bwd_graphed(self, x: torch.Tensor, y: torch.Tensor) -> None:
         if self._graph is None:
