@@ -67,6 +67,10 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
                         "default colocates a worker on rank 0")
     p.add_argument("--shrink-freq", type=int, default=50)
     p.add_argument("--checkpoint-freq", type=int, default=0)
+    p.add_argument("--wire-dtype", type=str, default="fp32",
+                   choices=["fp32", "bf16"],
+                   help="comm dtype: bf16 halves weight-push and svd-packet "
+                        "bytes over xGMI (fp32 default; invalid for qsgd)")
     p.add_argument("--resume", type=str, default=None, nargs="?", const="latest",
                    help="checkpoint path to resume from, or 'latest' to pick "
                         "the newest model_step_<N> in --train-dir")
@@ -134,6 +138,7 @@ class RunConfig:
             num_aggregate=a.num_aggregate,
             step_timeout=a.step_timeout,
             amp=a.amp,
+            wire_dtype=a.wire_dtype,
         )
 
 

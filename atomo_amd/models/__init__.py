@@ -31,8 +31,15 @@ _ZOO = {
     "VGG16": lambda num_classes=10, in_channels=3: vgg16_bn(num_classes, in_channels),
     "VGG19": lambda num_classes=10, in_channels=3: vgg19_bn(num_classes, in_channels),
     "AlexNet": lambda num_classes=10, in_channels=3: AlexNet(num_classes, in_channels),
+    # reference default config: depth 190 / growth 40
+    # (model_ops/densenet.py:60-116); "DenseNet40" keeps the light variant
     "DenseNet": lambda num_classes=10, in_channels=3: DenseNet(
-        num_classes=num_classes, in_channels=in_channels
+        depth=190, growth_rate=40, num_classes=num_classes,
+        in_channels=in_channels
+    ),
+    "DenseNet40": lambda num_classes=10, in_channels=3: DenseNet(
+        depth=40, growth_rate=12, num_classes=num_classes,
+        in_channels=in_channels
     ),
 }
 

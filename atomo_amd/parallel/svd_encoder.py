@@ -951,9 +951,16 @@ class BatchedSVDEncoder:
         device_sampled = use_kernels and self.codec.generator is None
         self.device_counted = device_sampled
         if device_sampled:
-            if not torch.cuda.is_current_stream_capturing():
+            if torch.cuda.is_current_stream_capturing():
+                # captured H2D copy: each graph replay re-reads the pinned
+                # scalar the trainer advances before replay
+                self.seed_dev.copy_(self.seed_host, non_blocking=True)
+            else:
+                # eager: bake this call's seed into a fill kernel — a
+                # non-blocking pinned copy would race (queued copies all
+                # read the latest host value)
                 self.advance_seed()
-            self.seed_dev.copy_(self.seed_host, non_blocking=True)
+                self.seed_dev.fill_(self._seed)
             e.sample_stage(
                 self.grams, self.evals_dev, self.stage_dev, self.desc,
                 self.eval_offs_dev, len(self.kernel_rows),

@@ -67,6 +67,7 @@ class PSTrainer:
         amp: bool = False,
         comm_type: str = "Bcast",
         num_aggregate: int = 0,
+        wire_dtype: str = "fp32",
     ):
         self.comm = comm
         self.device = device or comm.device
@@ -148,6 +149,38 @@ class PSTrainer:
             if self.is_master and not self.wc.reducible
             else None
         )
+        # opt-in bf16 WIRE: halves the bytes of the weight push and of the
+        # SVD factor packets over xGMI (justified by the
+        # N=8 comm phase).  fp32 stays the default — and the master copy,
+        # optimizer state and decode/apply all stay fp32.  Invalid for
+        # QSGD: its wire words are packed integer bit patterns that a
+        # dtype cast would destroy.
+        self.wire_bf16 = str(wire_dtype).lower() in ("bf16", "bfloat16")
+        if self.wire_bf16:
+            from ..codings import QSGDCodec
+
+            if isinstance(codec, QSGDCodec):
+                raise ValueError(
+                    "--wire-dtype bf16 is invalid for qsgd: its wire words "
+                    "are packed integer bit patterns"
+                )
+            # raw codec: only the weight push narrows; the gradient
+            # reduce stays fp32 (exact sums)
+            self._bcast_bf = torch.empty_like(self.flat, dtype=torch.bfloat16)
+            if comm.world > 1 and not self.wc.reducible:
+                self._wire_bf = torch.empty(
+                    self.wc.total_words, dtype=torch.bfloat16, device=self.device
+                )
+                self._gather_bf = (
+                    torch.empty(
+                        comm.world,
+                        self.wc.total_words,
+                        dtype=torch.bfloat16,
+                        device=self.device,
+                    )
+                    if self.is_master
+                    else None
+                )
         self.opt = (
             make_optimizer(
                 optimizer, self.flat, lr=lr, momentum=momentum, weight_decay=weight_decay
@@ -254,6 +287,12 @@ class PSTrainer:
                     self.comm.ps_push_weights(self.flat, self.step_num)
                 else:
                     self.comm.recv_weights(self.flat, src=0)
+            elif self.wire_bf16 and self.comm.world > 1:
+                if self.is_master:
+                    self._bcast_bf.copy_(self.flat)
+                self.comm.broadcast(self._bcast_bf, src=0)
+                if not self.is_master:
+                    self.flat.copy_(self._bcast_bf)
             else:
                 self.comm.broadcast(self.flat, src=0)
 
@@ -355,7 +394,13 @@ class PSTrainer:
                 if self.wc.reducible:
                     self.comm.reduce_sum(self.wire, dst=0)
                 elif self.comm.world > 1:
-                    self.comm.gather(self.wire, self.gather_buf, dst=0)
+                    if self.wire_bf16:
+                        self._wire_bf.copy_(self.wire)
+                        self.comm.gather(self._wire_bf, self._gather_bf, dst=0)
+                        if self.is_master:
+                            self.gather_buf.copy_(self._gather_bf)
+                    else:
+                        self.comm.gather(self.wire, self.gather_buf, dst=0)
 
             if self.is_master:
                 with t.phase("decode"):

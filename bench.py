@@ -50,6 +50,9 @@ def main():
     p.add_argument("--no-overlap", dest="overlap", action="store_false")
     p.add_argument("--amp", action="store_true", default=False,
                    help="bf16 autocast fwd/bwd (reported dtype changes)")
+    p.add_argument("--wire-dtype", type=str, default="fp32",
+                   choices=["fp32", "bf16"],
+                   help="comm dtype for weight push + svd factor packets")
     a = p.parse_args()
 
     from atomo_amd.codings import make_codec
@@ -85,6 +88,7 @@ def main():
         overlap=(a.code == "svd") if a.overlap is None else a.overlap,
         defer_loss=True,
         amp=a.amp,
+        wire_dtype=a.wire_dtype,
     )
     if a.channels_last:
         trainer.model.to(memory_format=torch.channels_last)

@@ -218,3 +218,63 @@ def test_dist_partial_aggregation_drops_straggler():
     slow_window = ps_durs[3:6]
     assert max(slow_window) < 0.3, ps_durs
     assert results[0][1] > 0  # stale packets were observed and dropped
+
+
+def _run_rank_bf16(rank, world, port, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.set_num_threads(2)
+    from atomo_amd.codings import make_codec
+    from atomo_amd.data import make_loaders
+    from atomo_amd.parallel import Comm, PSTrainer
+
+    comm = Comm(backend="gloo", device=torch.device("cpu"))
+    trainer = PSTrainer(
+        model_name="LeNet", codec=make_codec("svd", rank=3), comm=comm,
+        lr=0.05, momentum=0.9, num_classes=10, in_channels=1, seed=7,
+        device=torch.device("cpu"), wire_dtype="bf16",
+    )
+    train, _ = make_loaders("mnist", 16, 16, torch.device("cpu"), seed=50 + rank)
+    losses = [trainer.train_step(x, y) for _, (x, y) in zip(range(10), train)]
+    trainer.comm.broadcast(trainer.flat, src=0)
+    q.put((rank, losses, trainer.flat.sum().item()))
+    comm.barrier()
+    comm.close()
+
+
+def test_dist_bf16_wire():
+    """bf16 wire (weight push + svd packets): trains and keeps ranks
+    weight-synchronized."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_run_rank_bf16, args=(r, 2, 29641, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, losses, fsum = q.get(timeout=180)
+        results[rank] = (losses, fsum)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    import math
+
+    assert not any(math.isnan(l) for l in results[0][0])
+    assert results[0][0][-1] < results[0][0][0] * 1.5
+    assert results[0][1] == pytest.approx(results[1][1], rel=1e-5)
+
+
+def test_bf16_wire_rejects_qsgd():
+    from atomo_amd.codings import make_codec
+    from atomo_amd.parallel import Comm, PSTrainer
+
+    comm = Comm(device=torch.device("cpu"))
+    with pytest.raises(ValueError):
+        PSTrainer(
+            model_name="LeNet", codec=make_codec("qsgd"), comm=comm,
+            num_classes=10, in_channels=1, device=torch.device("cpu"),
+            wire_dtype="bf16",
+        )
