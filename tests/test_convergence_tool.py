@@ -1,6 +1,6 @@
-"""Teacher-student convergence benchmark: the task is non-memorizable
-(fresh random batches), so a decreasing loss is real learning and codec
-fidelity can visibly separate curves."""
+"""Gaussian-mixture convergence benchmark: the task is non-memorizable
+(fresh samples every step), so a decreasing loss is real learning and
+codec fidelity can visibly separate curves."""
 
 import json
 import sys
@@ -13,8 +13,8 @@ def test_convergence_tool_runs_and_learns(tmp_path):
     out = tmp_path / "conv.json"
     rc = main([
         "--network", "LeNet", "--dataset", "mnist", "--codes", "sgd,svd",
-        "--steps", "60", "--batch-size", "64", "--lr", "0.05",
-        "--eval-freq", "30", "--cpu", "--out", str(out),
+        "--steps", "100", "--batch-size", "64", "--lr", "0.02",
+        "--alpha", "0.3", "--eval-freq", "50", "--cpu", "--out", str(out),
     ])
     assert rc == 0
     data = json.loads(out.read_text())
@@ -25,4 +25,4 @@ def test_convergence_tool_runs_and_learns(tmp_path):
         tail = sum(losses[-10:]) / 10
         # fresh data every step: a falling loss cannot be memorization
         assert tail < head, (r["code"], head, tail)
-        assert r["agreement"][-1]["teacher_agreement"] > 0.2
+        assert r["holdout"][-1]["holdout_acc"] > 0.2
