@@ -228,6 +228,12 @@ class BatchedSVDEncoder:
             desc_rows, gram_work, sel_work, eval_offs = [], [], [], []
             rows_j64, rows_j128 = [], []
             self.solver_layers = []  # big folds solved by hipSOLVER
+            # layers whose (sm x r) selection tile exceeds the 160 KB LDS
+            # budget of batched_sel (e.g. AlexNet-227's 4096x9216 fc fold):
+            # they keep the device gram/solve/sample path but run their
+            # selection GEMM through rocBLAS with fixed r_max shapes
+            self.sel_mm_layers = []
+            SEL_LDS_CAP = 140 * 1024
             ev_off, sel_elems = 0, 1
             # group-level routing for big folds: batched hipSOLVER syevd is
             # nearly count-free per call, pooled host LAPACK wins for one or
@@ -282,7 +288,15 @@ class BatchedSVDEncoder:
                 ev_off += sm
                 self.kernel_set.add(i)
                 self.kernel_rows.append(i)
-                sel_elems = max(sel_elems, sm * min(R_CAP, s.meta["r_max"]))
+                sel_fit = (
+                    sm * min(R_CAP, s.meta["r_max"]) * 4 + 160 <= SEL_LDS_CAP
+                )
+                if sel_fit:
+                    sel_elems = max(
+                        sel_elems, sm * min(R_CAP, s.meta["r_max"])
+                    )
+                else:
+                    self.sel_mm_layers.append(i)
                 if sm <= SMALL_SM:
                     rows_j64.append(row)
                     for c in range((tall + GRAM_CHUNK - 1) // GRAM_CHUNK):
@@ -293,8 +307,9 @@ class BatchedSVDEncoder:
                 else:
                     self.solver_layers.append(i)
                     self.big_gram_layers.append(i)
-                for c in range((tall + SEL_CHUNK - 1) // SEL_CHUNK):
-                    sel_work.append([row, c])
+                if sel_fit:
+                    for c in range((tall + SEL_CHUNK - 1) // SEL_CHUNK):
+                        sel_work.append([row, c])
             if desc_rows:
                 dev = device
                 self.desc = torch.tensor(desc_rows, dtype=torch.int64, device=dev)
@@ -550,6 +565,52 @@ class BatchedSVDEncoder:
     # in _rsvd_tail_dev.  Atoms beyond the subspace read eval = 0 and are
     # never sampled.  (North-star "one-pass randomized SVD"; reference
     # semantics codings/svd.py:49-117.)
+    def _zero_oversize_stage(self) -> None:
+        """Zero the staged factors of sel-oversize layers: the sampler
+        writes only r_hat entries, and the fixed-r_max rocBLAS selection
+        below must see zeros beyond r_hat."""
+        for i in self.sel_mm_layers:
+            spec = self.specs[i]
+            sm = self.small[i]
+            r_max = spec.meta["r_max"]
+            so = self.stage_offsets[i]
+            seg = 1 + r_max * (1 + 2 * sm)
+            self.stage_dev[so : so + seg].zero_()
+
+    def _sel_oversize(self, flat_grad: torch.Tensor, wire: torch.Tensor):
+        """Selection GEMMs for layers whose (sm x r) tile exceeds the
+        batched_sel LDS budget (AlexNet-227 fc folds): fixed r_max shapes
+        straight from the device stage — no host sync; rows beyond the
+        sampled r_hat are zeros and decode ignores them."""
+        sd = self.stage_dev
+        for i in self.sel_mm_layers:
+            spec = self.specs[i]
+            m, n, r_max = spec.meta["m"], spec.meta["n"], spec.meta["r_max"]
+            sm = self.small[i]
+            wo, so = spec.wire_offset, self.stage_offsets[i]
+            wire[wo : wo + 1].copy_(sd[so : so + 1])  # r_hat header
+            wire[wo + 1 + r_max * m : wo + 1 + r_max * m + r_max].copy_(
+                sd[so + 1 : so + 1 + r_max]
+            )  # s_wire
+            facT = sd[so + 1 + r_max : so + 1 + r_max + r_max * sm].view(
+                r_max, sm
+            )
+            sel = sd[
+                so + 1 + r_max * (1 + sm) : so + 1 + r_max * (1 + sm)
+                + sm * r_max
+            ].view(sm, r_max)
+            a = flat_grad.narrow(0, self.param_offsets[i], m * n).view(m, n)
+            if self.m_is_tall[i]:
+                u_out = wire[wo + 1 : wo + 1 + r_max * m].view(r_max, m)
+                torch.mm(sel.t(), a.t(), out=u_out)
+                v_off = wo + 1 + r_max * (m + 1)
+                wire[v_off : v_off + r_max * n].view(r_max, n).copy_(facT)
+            else:
+                wire[wo + 1 : wo + 1 + r_max * m].view(r_max, m).copy_(facT)
+                v_off = wo + 1 + r_max * (m + 1)
+                v_out = wire[v_off : v_off + r_max * n].view(r_max, n)
+                torch.mm(sel.t(), a, out=v_out)
+
     def advance_seed(self) -> None:
         """Host-side LCG step for the on-device sampler seed.  Under a
         whole-step hipGraph the trainer calls this before each replay:
@@ -951,6 +1012,7 @@ class BatchedSVDEncoder:
         device_sampled = use_kernels and self.codec.generator is None
         self.device_counted = device_sampled
         if device_sampled:
+            self._zero_oversize_stage()
             if torch.cuda.is_current_stream_capturing():
                 # captured H2D copy: each graph replay re-reads the pinned
                 # scalar the trainer advances before replay
@@ -967,10 +1029,12 @@ class BatchedSVDEncoder:
                 self.codec.rank, not self.codec.random_sample, self.seed_dev,
                 self.used_words_dev,
             )
-            e.batched_sel(
-                flat_grad, wire, self.stage_dev, self.desc, self.sel_work,
-                self.sel_work.shape[0], self.sel_elems,
-            )
+            if self.sel_work.shape[0]:
+                e.batched_sel(
+                    flat_grad, wire, self.stage_dev, self.desc, self.sel_work,
+                    self.sel_work.shape[0], self.sel_elems,
+                )
+            self._sel_oversize(flat_grad, wire)
             if not host_layers:
                 mark("async sample+sel")
                 return -1  # used words accumulate in used_words_dev
@@ -1094,14 +1158,17 @@ class BatchedSVDEncoder:
 
         # ---- phase C: device stage build + batched sel + rocBLAS -------
         if use_kernels and not device_sampled:
+            self._zero_oversize_stage()
             e.build_stage(
                 self.grams, self.evals_dev, self.sel_table_dev, self.stage_dev,
                 self.desc, self.eval_offs_dev, len(self.kernel_rows),
             )
-            e.batched_sel(
-                flat_grad, wire, self.stage_dev, self.desc, self.sel_work,
-                self.sel_work.shape[0], self.sel_elems,
-            )
+            if self.sel_work.shape[0]:
+                e.batched_sel(
+                    flat_grad, wire, self.stage_dev, self.desc, self.sel_work,
+                    self.sel_work.shape[0], self.sel_elems,
+                )
+            self._sel_oversize(flat_grad, wire)
         sd = self.stage_dev
         for i, r_hat in host_plans:
             spec = specs[i]

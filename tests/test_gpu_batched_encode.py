@@ -225,3 +225,30 @@ def test_randomized_matches_exact_eigenvalues(dev):
         ev_x = (s[:r_max] ** 2)
         rel = ((ev_r - ev_x).abs() / ev_x.clamp(min=1e-12)).max()
         assert rel < 1e-2, (spec.shape, float(rel))
+
+
+def test_sel_oversize_layer_roundtrip(dev):
+    """Layers whose (sm x r) selection tile exceeds the batched_sel LDS
+    budget (AlexNet-227 fc folds) take the rocBLAS fixed-r_max selection
+    path; truncation mode must still reproduce the top-r SVD."""
+    torch.manual_seed(4)
+    shapes = [(3000, 2816), (64, 16, 3, 3)]
+    codec, specs, enc, flat, grads, wire = _build(dev, shapes, rank=3,
+                                                  exact_eigh=False)
+    assert enc.sel_mm_layers, "big fc fold should route to rocBLAS selection"
+    _decaying_grads(flat, grads, specs, decay=8.0)
+    for _ in range(4):
+        enc.encode_all(grads, wire, flat_grad=flat)
+    torch.cuda.synchronize()
+    from atomo_amd.codings.svd import grad_to_2d
+
+    for g, spec in zip(grads, specs):
+        region = wire[spec.wire_offset : spec.wire_offset + spec.wire_words]
+        out = torch.zeros(spec.numel, device=dev)
+        codec.decode_from(region, out, spec)
+        a = grad_to_2d(g)
+        u, s, vh = torch.linalg.svd(a, full_matrices=False)
+        r = min(codec.rank, spec.meta["r_max"])
+        best = ((u[:, :r] * s[:r]) @ vh[:r]).reshape(-1)[: spec.numel]
+        rel = (out - best).norm() / best.norm()
+        assert rel < 2e-2, (spec.shape, float(rel))
