@@ -37,10 +37,10 @@ void atomo_batched_gram_launch(const float*, float*, const int64_t*,
 void atomo_batched_sel_launch(const float*, float*, const float*,
                               const int64_t*, const int32_t*, int, int,
                               hipStream_t);
-void atomo_jacobi_dense_launch(float*, float*, int, int, hipStream_t);
+void atomo_jacobi_dense_launch(float*, float*, int, int, int, hipStream_t);
 void atomo_jacobi_eigh_launch(float*, float*, const int64_t*, const int64_t*,
                               const int32_t*, int, int, float*,
-                              const int64_t*, int, hipStream_t);
+                              const int64_t*, int, int, hipStream_t);
 void atomo_jacobi_eigh_big_launch(float*, float*, float*, const int64_t*,
                                   const int64_t*, const int32_t*,
                                   const int64_t*, int, hipStream_t);
@@ -270,7 +270,7 @@ void batched_sel(torch::Tensor flat, torch::Tensor wire, torch::Tensor stage,
 void jacobi_eigh(torch::Tensor grams, torch::Tensor evals, torch::Tensor desc,
                  torch::Tensor eval_offs, torch::Tensor rows, int64_t n_mats,
                  int64_t jmax, torch::Tensor vwarm, torch::Tensor vwarm_offs,
-                 int64_t warm) {
+                 int64_t warm, int64_t max_sweeps) {
   check_f32_cuda(grams, "grams");
   check_f32_cuda(evals, "evals");
   TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == torch::kInt64 &&
@@ -287,11 +287,11 @@ void jacobi_eigh(torch::Tensor grams, torch::Tensor evals, torch::Tensor desc,
       rows.data_ptr<int32_t>(), (int)n_mats, (int)jmax,
       vwarm.numel() > 1 ? vwarm.data_ptr<float>() : nullptr,
       vwarm_offs.numel() ? vwarm_offs.data_ptr<int64_t>() : nullptr,
-      (int)warm, cur_stream());
+      (int)warm, (int)max_sweeps, cur_stream());
 }
 
 void jacobi_dense(torch::Tensor a, torch::Tensor evals, int64_t n_mats,
-                  int64_t nb) {
+                  int64_t nb, int64_t max_sweeps) {
   check_f32_cuda(a, "a");
   check_f32_cuda(evals, "evals");
   TORCH_CHECK(nb <= 64 && nb % 2 == 0, "nb must be even and <= 64");
@@ -299,7 +299,8 @@ void jacobi_dense(torch::Tensor a, torch::Tensor evals, int64_t n_mats,
               "jacobi_dense: buffer too small");
   if (n_mats == 0) return;
   atomo_jacobi_dense_launch(a.data_ptr<float>(), evals.data_ptr<float>(),
-                            (int)n_mats, (int)nb, cur_stream());
+                            (int)n_mats, (int)nb, (int)max_sweeps,
+                            cur_stream());
 }
 
 void jacobi_eigh_big(torch::Tensor grams, torch::Tensor vbuf,
@@ -362,11 +363,15 @@ void sample_stage(torch::Tensor evecs, torch::Tensor evals,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sample_stage", &sample_stage,
         "fused on-device Bernoulli atom sampler + stage builder");
-  m.def("jacobi_eigh", &jacobi_eigh,
+  m.def("jacobi_eigh", &jacobi_eigh, py::arg("grams"), py::arg("evals"),
+        py::arg("desc"), py::arg("eval_offs"), py::arg("rows"),
+        py::arg("n_mats"), py::arg("jmax"), py::arg("vwarm"),
+        py::arg("vwarm_offs"), py::arg("warm"), py::arg("max_sweeps") = 10,
         "batched parallel-Jacobi symmetric eigensolver (sm <= 64, LDS)");
   m.def("jacobi_eigh_big", &jacobi_eigh_big,
         "batched parallel-Jacobi eigensolver (64 < sm <= 512, L2-resident)");
-  m.def("jacobi_dense", &jacobi_dense,
+  m.def("jacobi_dense", &jacobi_dense, py::arg("a"), py::arg("evals"),
+        py::arg("n_mats"), py::arg("nb"), py::arg("max_sweeps") = 10,
         "batched dense small (nb<=64) symmetric eigh, one wave per matrix; "
         "eigenvalues descending, eigenvectors overwrite the input");
   m.def("build_stage", &build_stage,

@@ -66,7 +66,7 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
     const int64_t* __restrict__ desc, const int64_t* __restrict__ eval_offs,
     const int32_t* __restrict__ rows_list, int n_mats,
     float* __restrict__ vwarm, const int64_t* __restrict__ vwarm_offs,
-    int save_warm) {
+    int save_warm, int max_sweeps) {
   constexpr int JSTRIDE = JMAX + 1;
   __shared__ float G[JMAX * JSTRIDE];
   __shared__ float V[JMAX * JSTRIDE];
@@ -130,7 +130,7 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
   const float tol2 = fro_all * 1e-10f;
 
   const int np = N / 2;
-  for (int sweep = 0; sweep < SWEEPS && !done_s; ++sweep) {
+  for (int sweep = 0; sweep < max_sweeps && !done_s; ++sweep) {
     for (int round = 0; round < N - 1; ++round) {
       if (tid < np) {
         auto player = [&](int slot) {
@@ -229,7 +229,8 @@ __global__ void __launch_bounds__(JTHREADS) jacobi_eigh_kernel(
 #define DTHREADS 64
 
 __global__ void __launch_bounds__(DTHREADS) jacobi_dense_kernel(
-    float* __restrict__ a, float* __restrict__ evals, int n_mats, int nb) {
+    float* __restrict__ a, float* __restrict__ evals, int n_mats, int nb,
+    int max_sweeps) {
   constexpr int DSTRIDE = DMAX + 1;
   __shared__ float G[DMAX * DSTRIDE];
   __shared__ float V[DMAX * DSTRIDE];
@@ -262,7 +263,7 @@ __global__ void __launch_bounds__(DTHREADS) jacobi_dense_kernel(
   const float tol2 = fro_s * 1e-10f;
 
   const int np = N / 2;
-  for (int sweep = 0; sweep < SWEEPS && !done_s; ++sweep) {
+  for (int sweep = 0; sweep < max_sweeps && !done_s; ++sweep) {
     for (int round = 0; round < N - 1; ++round) {
       if (tid < np) {
         auto player = [&](int slot) {
@@ -659,7 +660,7 @@ void atomo_jacobi_eigh_launch(float* grams, float* evals, const int64_t* desc,
                               const int64_t* eval_offs,
                               const int32_t* rows_list, int n_mats, int jmax,
                               float* vwarm, const int64_t* vwarm_offs,
-                              int warm, hipStream_t stream) {
+                              int warm, int max_sweeps, hipStream_t stream) {
   // warm < 0: no warm basis at all (also skip the save)
   const int save = (warm >= 0 && vwarm != nullptr) ? 1 : 0;
   if (jmax <= 64) {
@@ -667,16 +668,17 @@ void atomo_jacobi_eigh_launch(float* grams, float* evals, const int64_t* desc,
       hipLaunchKernelGGL((jacobi_eigh_kernel<64, true>), dim3(n_mats),
                          dim3(JTHREADS), 0, stream, grams, evals, desc,
                          eval_offs, rows_list, n_mats, vwarm, vwarm_offs,
-                         save);
+                         save, max_sweeps);
     else
       hipLaunchKernelGGL((jacobi_eigh_kernel<64, false>), dim3(n_mats),
                          dim3(JTHREADS), 0, stream, grams, evals, desc,
                          eval_offs, rows_list, n_mats, vwarm, vwarm_offs,
-                         save);
+                         save, max_sweeps);
   } else {
     hipLaunchKernelGGL((jacobi_eigh_kernel<128, false>), dim3(n_mats),
                        dim3(JTHREADS), 0, stream, grams, evals, desc,
-                       eval_offs, rows_list, n_mats, vwarm, vwarm_offs, 0);
+                       eval_offs, rows_list, n_mats, vwarm, vwarm_offs, 0,
+                       max_sweeps);
   }
 }
 
@@ -692,9 +694,9 @@ void atomo_jacobi_eigh_big_launch(float* grams, float* vbuf, float* evals,
 }
 
 void atomo_jacobi_dense_launch(float* a, float* evals, int n_mats, int nb,
-                               hipStream_t stream) {
+                               int max_sweeps, hipStream_t stream) {
   hipLaunchKernelGGL(jacobi_dense_kernel, dim3(n_mats), dim3(DTHREADS), 0,
-                     stream, a, evals, n_mats, nb);
+                     stream, a, evals, n_mats, nb, max_sweeps);
 }
 
 void atomo_build_stage_launch(const float* evecs, const float* evals,
