@@ -473,6 +473,7 @@ __global__ void __launch_bounds__(JBIG_THREADS) jacobi_eigh_big_kernel(
 // stage builder: per layer gather the sampled atoms into the wire staging
 // sel table row (fp32): [r_hat | idx 0..31 (as float) | probs 0..31]
 // stage layout:          [r_hat | s_wire(r_max) | facT(r_max*sm) | sel(sm*r_max)]
+// facT rows / sel cols use FIXED r_max stride, zero beyond r_hat
 // ---------------------------------------------------------------------------
 __global__ void __launch_bounds__(64) build_stage_kernel(
     const float* __restrict__ evecs, const float* __restrict__ evals,
@@ -504,14 +505,15 @@ __global__ void __launch_bounds__(64) build_stage_kernel(
     inv_s[tid] = s > 1e-12f ? 1.f / s : 0.f;
     stage[so + 1 + tid] = s_sel[tid];
   }
+  if (tid < r_max && tid >= r_hat) stage[so + 1 + tid] = 0.f;
   __syncthreads();
-  float* facT = stage + so + 1 + r_max;                     // (r_hat, sm)
-  float* sel = stage + so + 1 + (int64_t)r_max * (1 + sm);  // (sm, r_hat)
-  for (int i = tid; i < r_hat * sm; i += 64) {
+  float* facT = stage + so + 1 + r_max;                     // (r_max, sm)
+  float* sel = stage + so + 1 + (int64_t)r_max * (1 + sm);  // (sm, r_max)
+  for (int i = tid; i < r_max * sm; i += 64) {
     const int r = i / sm, k = i % sm;
-    const float v = V[(int64_t)k * sm + idxs[r]];
+    const float v = (r < r_hat) ? V[(int64_t)k * sm + idxs[r]] : 0.f;
     facT[(int64_t)r * sm + k] = v;
-    sel[(int64_t)k * r_hat + r] = v * inv_s[r];
+    sel[(int64_t)k * r_max + r] = (r < r_hat) ? v * inv_s[r] : 0.f;
   }
 }
 
@@ -627,14 +629,15 @@ __global__ void __launch_bounds__(64) sample_stage_kernel(
   }
   __syncthreads();
   const int r_hat = r_hat_s;
-  for (int k = tid; k < r_hat; k += 64) stage[so + 1 + k] = s_sel[k];
+  for (int k = tid; k < r_max; k += 64)
+    stage[so + 1 + k] = (k < r_hat) ? s_sel[k] : 0.f;
   float* facT = stage + so + 1 + r_max;
   float* sel = stage + so + 1 + (int64_t)r_max * (1 + sm);
-  for (int i = tid; i < r_hat * sm; i += 64) {
+  for (int i = tid; i < r_max * sm; i += 64) {
     const int r = i / sm, k = i % sm;
-    const float v = V[(int64_t)k * sm + idxs[r]];
+    const float v = (r < r_hat) ? V[(int64_t)k * sm + idxs[r]] : 0.f;
     facT[(int64_t)r * sm + k] = v;
-    sel[(int64_t)k * r_hat + r] = v * inv_s[r];
+    sel[(int64_t)k * r_max + r] = (r < r_hat) ? v * inv_s[r] : 0.f;
   }
 }
 

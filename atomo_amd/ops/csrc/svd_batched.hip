@@ -104,7 +104,7 @@ __device__ __forceinline__ void sel_rows(const float* __restrict__ A,
                                          float* __restrict__ wire,
                                          const float* __restrict__ sel_lds,
                                          int t_begin, int tall, int m, int n,
-                                         bool is_tall, int r_hat,
+                                         bool is_tall, int r_hat, int r_max,
                                          int64_t t_dst) {
   const int sm = is_tall ? n : m;
   for (int rr = 0; rr < SEL_ROWS_PER_THREAD; ++rr) {
@@ -117,7 +117,7 @@ __device__ __forceinline__ void sel_rows(const float* __restrict__ A,
       const float* a_row = A + (int64_t)t * n;
       for (int s = 0; s < sm; ++s) {
         const float av = a_row[s];
-        const float* sl = sel_lds + (int64_t)s * r_hat;
+        const float* sl = sel_lds + (int64_t)s * r_max;
 #pragma unroll
         for (int r = 0; r < RCAP; ++r)
           acc[r] = fmaf(av, (r < r_hat) ? sl[r] : 0.f, acc[r]);
@@ -125,7 +125,7 @@ __device__ __forceinline__ void sel_rows(const float* __restrict__ A,
     } else {
       for (int s = 0; s < sm; ++s) {
         const float av = A[(int64_t)s * n + t];
-        const float* sl = sel_lds + (int64_t)s * r_hat;
+        const float* sl = sel_lds + (int64_t)s * r_max;
 #pragma unroll
         for (int r = 0; r < RCAP; ++r)
           acc[r] = fmaf(av, (r < r_hat) ? sl[r] : 0.f, acc[r]);
@@ -161,8 +161,8 @@ __global__ void __launch_bounds__(256) batched_sel_kernel(
     if (cached_layer_s != layer) {
       __syncthreads();
       const float* sel = stage + so + 1 + (int64_t)r_max * (1 + sm);
-      for (int i = threadIdx.x; i < sm * r_hat; i += blockDim.x)
-        sel_lds[i] = sel[i];  // sel[s*r_hat... stored (sm, r_hat) row-major
+      for (int i = threadIdx.x; i < sm * r_max; i += blockDim.x)
+        sel_lds[i] = sel[i];  // stored (sm, r_max) row-major, 0 past r_hat
       if (threadIdx.x == 0) cached_layer_s = layer;
       __syncthreads();
     }
@@ -185,15 +185,17 @@ __global__ void __launch_bounds__(256) batched_sel_kernel(
     const int64_t t_dst = is_tall ? wo + 1 : wo + 1 + (int64_t)r_max * (m + 1);
     const int t_begin = chunk * SEL_CHUNK;
     if (r_hat <= 4)
-      sel_rows<4>(A, wire, sel_lds, t_begin, tall, m, n, is_tall, r_hat, t_dst);
+      sel_rows<4>(A, wire, sel_lds, t_begin, tall, m, n, is_tall, r_hat,
+                  r_max, t_dst);
     else if (r_hat <= 8)
-      sel_rows<8>(A, wire, sel_lds, t_begin, tall, m, n, is_tall, r_hat, t_dst);
+      sel_rows<8>(A, wire, sel_lds, t_begin, tall, m, n, is_tall, r_hat,
+                  r_max, t_dst);
     else if (r_hat <= 16)
       sel_rows<16>(A, wire, sel_lds, t_begin, tall, m, n, is_tall, r_hat,
-                   t_dst);
+                   r_max, t_dst);
     else
       sel_rows<32>(A, wire, sel_lds, t_begin, tall, m, n, is_tall, r_hat,
-                   t_dst);
+                   r_max, t_dst);
   }
 #undef cached_layer_s
 }
