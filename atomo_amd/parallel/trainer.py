@@ -241,6 +241,11 @@ class PSTrainer:
                 )
         # backward-hook overlap (reference *Split capability): per-layer
         # encode work on a side stream while backward continues.
+        if self.use_graph and not self.graph_whole and comm.world == 1:
+            # whole-step capture ineligible (host layers / pinned RNG /
+            # qsgd): backward-hook overlap beats a fwd/bwd-only graph for
+            # the svd encode path, so let overlap win
+            self.use_graph = False
         self.overlap = (
             bool(overlap)
             and self.device.type == "cuda"

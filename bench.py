@@ -40,8 +40,10 @@ def main():
     p.add_argument("--cpu", action="store_true", default=False)
     p.add_argument("--phase-log", action="store_true", default=False,
                    help="print per-phase timer breakdown to stderr")
-    p.add_argument("--graph", action="store_true", default=False,
-                   help="capture forward/backward in a hipGraph and replay")
+    p.add_argument("--graph", action="store_true", default=None,
+                   help="hipGraph capture (whole step at N=1 when the path "
+                        "is pure device kernels; default: on for N==1)")
+    p.add_argument("--no-graph", dest="graph", action="store_false")
     p.add_argument("--channels-last", action="store_true", default=False,
                    help="NHWC memory format for convs")
     p.add_argument("--overlap", action="store_true", default=None,
@@ -84,7 +86,7 @@ def main():
         dedicated_ps=a.dedicated_ps,
         seed=42,
         device=device,
-        use_graph=a.graph,
+        use_graph=(comm.world == 1) if a.graph is None else a.graph,
         overlap=(a.code == "svd") if a.overlap is None else a.overlap,
         defer_loss=True,
         amp=a.amp,

@@ -64,6 +64,38 @@ def main(out_dir, dest):
                 ent["dispatches"] += len(disp.get(k, ()))
                 for cn, v in cs.items():
                     ent["sums"][cn] = ent["sums"].get(cn, 0.0) + v
+        elif "kernel_trace" in base:
+            # per-dispatch timestamps: steady-state stats from the second
+            # half of the trace window (excludes MIOpen find-mode warmup)
+            byk = defaultdict(list)
+            t0, t1 = None, None
+            for d in rows:
+                k = kname(d)
+                try:
+                    s0 = float(d.get("start_timestamp") or d.get("start") or 0)
+                    e0 = float(d.get("end_timestamp") or d.get("end") or 0)
+                except (TypeError, ValueError):
+                    continue
+                if not k or e0 <= s0:
+                    continue
+                byk[k].append((s0, e0 - s0))
+                t0 = s0 if t0 is None else min(t0, s0)
+                t1 = e0 if t1 is None else max(t1, e0)
+            if t0 is None:
+                continue
+            mid = t0 + 0.5 * (t1 - t0)
+            ss = summary.setdefault("steady_state", {})
+            for k, evs in byk.items():
+                late = sorted(dur for s0, dur in evs if s0 >= mid)
+                if not late:
+                    continue
+                n = len(late)
+                ss[k] = {
+                    "dispatches": n,
+                    "total_us": sum(late) / 1e3,
+                    "p50_us": late[n // 2] / 1e3,
+                    "p95_us": late[min(n - 1, int(n * 0.95))] / 1e3,
+                }
         elif "kernel_stats" in base or ("stats" in base and "domain" not in base):
             for d in rows:
                 k = kname(d)
