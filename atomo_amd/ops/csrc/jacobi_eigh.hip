@@ -32,7 +32,11 @@
 
 #define GD_N 8
 #define SWEEPS 10
-#define JTHREADS 256
+// one wave per matrix: the kernel is one-WG-per-matrix and its runtime is
+// the per-round LDS/barrier latency chain, not throughput — 4-wave WGs
+// paid ~3 full barriers per rotation round (measured 659 us for a warm
+// 3-sweep call); single-wave barriers are nearly free
+#define JTHREADS 64
 #define JBIG_THREADS 512
 #define R_CAP 32
 #define SEL_ROW 65  // [r_hat | idx*32 | probs*32]

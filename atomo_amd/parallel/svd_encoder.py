@@ -664,7 +664,7 @@ class BatchedSVDEncoder:
         s = torch.cat(
             [torch.bmm(y.transpose(1, 2), y) for y in ys], dim=0
         ).contiguous()
-        lam, v = self._dense_eigh(s, sweeps=5)
+        lam, v = self._dense_eigh(s, sweeps=3)
         inv = lam.clamp(min=1e-6).rsqrt()
         row0, out = 0, []
         for y in ys:
@@ -710,7 +710,7 @@ class BatchedSVDEncoder:
             ],
             dim=0,
         ).contiguous()
-        lam_all, w_all = self._dense_eigh(ts, sweeps=6)
+        lam_all, w_all = self._dense_eigh(ts, sweeps=5)
         lam_all = lam_all.clamp(min=0.0)
         row0 = 0
         for (sm, idxs, gather, scatter, evi, tails), q, tr in zip(
