@@ -32,11 +32,10 @@
 
 #define GD_N 8
 #define SWEEPS 10
-// one wave per matrix: the kernel is one-WG-per-matrix and its runtime is
-// the per-round LDS/barrier latency chain, not throughput — 4-wave WGs
-// paid ~3 full barriers per rotation round (measured 659 us for a warm
-// 3-sweep call); single-wave barriers are nearly free
-#define JTHREADS 64
+// 256 threads/WG: measured faster than single-wave — the warm
+// pre-rotation (two N^3 LDS matmuls) is throughput-bound and wants the
+// threads; the dense small-batch variant below is the opposite case
+#define JTHREADS 256
 #define JBIG_THREADS 512
 #define R_CAP 32
 #define SEL_ROW 65  // [r_hat | idx*32 | probs*32]

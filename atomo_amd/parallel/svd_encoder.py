@@ -984,7 +984,7 @@ class BatchedSVDEncoder:
                     self.grams, self.evals_dev, self.desc, self.eval_offs_dev,
                     self.rows_j64, self.rows_j64.shape[0], 64,
                     self.vwarm, self.vwarm_offs, 1 if self._warm else 0,
-                    3 if self._warm else 8,
+                    2 if self._warm else 8,
                 )
                 self._warm = True
                 if self.rows_j128.shape[0]:
