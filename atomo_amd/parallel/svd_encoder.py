@@ -167,6 +167,7 @@ class BatchedSVDEncoder:
         self.specs = list(specs)
         self.param_offsets = param_offsets
         self._pool = ThreadPoolExecutor(max_workers=8)
+        self._jac_stream = None  # side stream for the LDS Jacobi fork/join
         # MKL builds a fresh OpenMP team the first time a NEW thread calls
         # LAPACK (~20 ms each); warm every pool worker once so in-step host
         # eigensolves run at their true sub-ms cost.
@@ -976,28 +977,39 @@ class BatchedSVDEncoder:
                     grams_host = self.grams_host
                     gram_event = torch.cuda.Event()
                     gram_event.record()
-                # warm steps: the pre-rotated Gram is near-diagonal, and
-                # V stays orthonormal at any sweep cap (see _dense_eigh
-                # doc) — 3 sweeps replace a measured ~1.3 ms of full
-                # sweeps with ~0.4 ms
-                e.jacobi_eigh(
-                    self.grams, self.evals_dev, self.desc, self.eval_offs_dev,
-                    self.rows_j64, self.rows_j64.shape[0], 64,
-                    self.vwarm, self.vwarm_offs, 1 if self._warm else 0,
-                    2 if self._warm else 8,
-                )
-                self._warm = True
-                if self.rows_j128.shape[0]:
+                # LDS Jacobi on a SIDE stream, concurrent with the big-fold
+                # solve: jacobi_eigh's runtime is one matrix's critical path
+                # on ONE CU (62 workgroups leave 190+ CUs idle), while the
+                # randomized solver is chip-wide rocBLAS GEMMs — they touch
+                # disjoint gram/eval slots, so fork/join events overlap them
+                # (also inside a whole-step graph capture: event fork/join
+                # is the supported multi-stream capture pattern).
+                # Warm steps run 2 sweeps: the pre-rotated Gram is
+                # near-diagonal and V stays orthonormal at any sweep cap
+                # (see _dense_eigh doc).
+                if self._jac_stream is None:
+                    self._jac_stream = torch.cuda.Stream()
+                fork = torch.cuda.Event()
+                fork.record()
+                with torch.cuda.stream(self._jac_stream):
+                    self._jac_stream.wait_event(fork)
                     e.jacobi_eigh(
                         self.grams, self.evals_dev, self.desc,
-                        self.eval_offs_dev, self.rows_j128,
-                        self.rows_j128.shape[0], 128,
-                        self.vwarm, self.vwarm_offs, -1, 8,
+                        self.eval_offs_dev,
+                        self.rows_j64, self.rows_j64.shape[0], 64,
+                        self.vwarm, self.vwarm_offs, 1 if self._warm else 0,
+                        2 if self._warm else 8,
                     )
-                # big folds: batched hipSOLVER syevd per size-group, each
-                # group on its own stream so independent solves overlap
-                # (serially they cost e.g. 5.6+11.5+21 ms on ResNet-50);
-                # results written back into the gram slots / evals buffer
+                    if self.rows_j128.shape[0]:
+                        e.jacobi_eigh(
+                            self.grams, self.evals_dev, self.desc,
+                            self.eval_offs_dev, self.rows_j128,
+                            self.rows_j128.shape[0], 128,
+                            self.vwarm, self.vwarm_offs, -1, 8,
+                        )
+                    join = torch.cuda.Event()
+                    join.record(self._jac_stream)
+                self._warm = True
                 if self.solver_layers and not (
                     grams_done and getattr(self, "_ov_plan", None)
                 ):
@@ -1009,6 +1021,7 @@ class BatchedSVDEncoder:
                     else:
                         self._solve_big_folds_exact()
                     mark("A3 big-fold solve")
+                torch.cuda.current_stream().wait_event(join)
                 self.evals_host.copy_(self.evals_dev, non_blocking=True)
             elif host_layers:
                 grams_host = self.grams.to("cpu")  # synchronous copy
