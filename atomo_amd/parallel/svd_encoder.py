@@ -62,7 +62,10 @@ SMALL_SM = 64      # LDS-Jacobi (64-variant) + batched_gram kernel
 J128_SM = int(os.environ.get("ATOMO_JACOBI_CAP", "64"))
 SOLVER_SM = 4096
 
-# measured per-matrix costs (ms) on MI355X + EPYC host, fp32
+# measured per-matrix costs (ms) on MI355X + EPYC host, fp32 — used ONLY
+# to route the ATOMO_EXACT_EIGH=1 oracle (the default randomized solver
+# device-routes every big fold, so these tables no longer gate the hot
+# path; re-measure if the oracle's host/solver ratio matters on new HW)
 _HOST_EIGH_MS = {128: 0.9, 256: 3.3, 512: 10.7, 1024: 45.0, 2048: 160.0}
 _SOLVER_EIGH_MS = {128: 2.1, 256: 5.6, 512: 11.5, 1024: 21.0, 2048: 44.0}
 # hipSOLVER syevd cost model: per-call base + per-extra-matrix increment

@@ -343,7 +343,14 @@ class PSTrainer:
                 if self.num_aggregate > 0:
                     # TRUE partial aggregation: the PS returns after the
                     # first K contributions; a straggler's packet is left
-                    # pending and dropped (stale) when it lands next step
+                    # pending and dropped (stale) when it lands next step.
+                    # The colocated PS's own contribution is always in the
+                    # first-K set — that is physical (zero latency), the
+                    # same property the reference's waitany drain would
+                    # have; which K-1 workers fill the rest is genuine
+                    # arrival order, not rank order (ADVICE r1 fairness
+                    # note: with a persistent straggler the EXCLUDED set
+                    # is whoever is actually late, not a fixed rank)
                     contrib = [0]
                     if self.is_master:
                         self.agg.zero_()
