@@ -228,6 +228,9 @@ class PSTrainer:
         self._wgraph = None
         self._wgraph_lr = None
         self.graph_whole = False
+        if self.use_graph and self.amp:
+            # the captured bodies run fp32; don't silently drop autocast
+            self.use_graph = False
         if self.use_graph and comm.world == 1 and not self.dedicated_ps:
             if self.wc.reducible:
                 self.graph_whole = True
@@ -241,10 +244,36 @@ class PSTrainer:
                 )
         # backward-hook overlap (reference *Split capability): per-layer
         # encode work on a side stream while backward continues.
-        if self.use_graph and not self.graph_whole and comm.world == 1:
-            # whole-step capture ineligible (host layers / pinned RNG /
-            # qsgd): backward-hook overlap beats a fwd/bwd-only graph for
-            # the svd encode path, so let overlap win
+        # SPLIT graphs at world>1 (collective-gather mode, colocated):
+        # capture (zero+fwd/bwd+encode) and, on the PS, (decode+apply) as
+        # two graphs with the eager RCCL gather/broadcast between replays
+        # — recovers the whole-step graph's launch-latency win at N>1
+        # without capturing collectives.
+        self.graph_split = False
+        if (
+            self.use_graph
+            and comm.world > 1
+            and not self.dedicated_ps
+            and not self.p2p
+            and not self.wire_bf16
+        ):
+            if self.wc.reducible:
+                self.graph_split = True
+            else:
+                enc = self.wc._batched_encoder
+                self.graph_split = bool(
+                    enc is not None
+                    and enc.use_kernels
+                    and len(enc.kernel_set) == len(self.wc.specs)
+                    and self.codec.generator is None
+                )
+        self._graphA = None
+        self._graphB = None
+        self._graphB_lr = None
+        if self.use_graph and not self.graph_whole and not self.graph_split:
+            # capture ineligible (host layers / pinned RNG / qsgd):
+            # backward-hook overlap beats a fwd/bwd-only graph for the
+            # svd encode path, so let overlap win
             self.use_graph = False
         self.overlap = (
             bool(overlap)
@@ -283,6 +312,10 @@ class PSTrainer:
                 ):
                     self.save_checkpoint()
                 return self.last_loss
+        if self.graph_split and self.step_num >= 1:
+            # step 0 runs the standard eager path so every decode/apply
+            # kernel is warm before graph B captures
+            return self._train_step_split(x, y)
         with t.phase("fetch"):
             if self.p2p and self.num_aggregate > 0:
                 # partial mode: per-worker pipelined weight push — a
@@ -450,6 +483,127 @@ class PSTrainer:
     def current_loss(self) -> float:
         if self._loss_tensor is not None:
             self.last_loss = float(self._loss_tensor)
+        return self.last_loss
+
+    def _post_step(self) -> None:
+        self.step_num += 1
+        if self.watchdog is not None:
+            self.watchdog.step()
+        if self.step_num % self.shrink_freq == 0:
+            self.lr *= self.lr_shrinkage
+        if (
+            self.is_master
+            and self.checkpoint_freq > 0
+            and self.step_num % self.checkpoint_freq == 0
+        ):
+            self.save_checkpoint()
+
+    def _body_A(self) -> None:
+        """Pre-comm half of a split-graph step: zero + fwd/bwd + encode
+        on static buffers (no optimizer mutation — safe to warm-run)."""
+        self.flat_grad.zero_()
+        loss = self.loss_fn(self.model(self._static_x), self._static_y)
+        loss.backward()
+        self._static_loss = loss.detach()
+        if not self.wc.reducible:
+            self.wc.encode_all(self.wire, flat_grad=self.flat_grad)
+
+    def _graphA_step(self, x: torch.Tensor, y: torch.Tensor) -> bool:
+        if self._graphA is None:
+            try:
+                self._static_x = x.clone()
+                self._static_y = y.clone()
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    for _ in range(3):
+                        self._body_A()
+                torch.cuda.current_stream().wait_stream(side)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._body_A()
+                self._graphA = g
+            except Exception as exc:
+                print(
+                    f"[atomo] split-graph A capture failed ({exc}); eager",
+                    flush=True,
+                )
+                self.graph_split = False
+                return False
+        self._static_x.copy_(x)
+        self._static_y.copy_(y)
+        enc = self.wc._batched_encoder
+        if enc is not None:
+            enc.advance_seed()
+        self._graphA.replay()
+        if self.defer_loss:
+            self._loss_tensor = self._static_loss
+        else:
+            self.last_loss = float(self._static_loss)
+        return True
+
+    def _graphB_step(self) -> bool:
+        """PS-side decode+apply graph (capture needs kernels warmed by an
+        eager step 0; recaptures when lr changes)."""
+        if self._graphB is None or self._graphB_lr != self.lr:
+            try:
+                self.opt.lr = self.lr
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    if self.wc.reducible:
+                        self._apply(self.flat_grad)
+                    else:
+                        self.agg.zero_()
+                        self.wc.decode_all(self.gather_buf, self.agg)
+                        self._apply(self.agg)
+                self._graphB = g
+                self._graphB_lr = self.lr
+            except Exception as exc:
+                print(
+                    f"[atomo] split-graph B capture failed ({exc}); eager",
+                    flush=True,
+                )
+                self._graphB = None
+                self.graph_split = False
+                return False
+        self._graphB.replay()
+        return True
+
+    def _train_step_split(self, x: torch.Tensor, y: torch.Tensor) -> float:
+        t = self.timers
+        with t.phase("fetch"):
+            self.comm.broadcast(self.flat, src=0)
+        with t.phase("comp"):
+            ok = self._graphA_step(x, y)
+            if not ok:  # capture failed (graph_split now False): eager body
+                self.flat_grad.zero_()
+                loss = self.loss_fn(self.model(x), y)
+                loss.backward()
+                if self.defer_loss:
+                    self._loss_tensor = loss.detach()
+                else:
+                    self.last_loss = float(loss.detach())
+                if not self.wc.reducible:
+                    self.wc.encode_all(self.wire, flat_grad=self.flat_grad)
+        with t.phase("comm"):
+            if self.wc.reducible:
+                self.comm.reduce_sum(self.wire, dst=0)
+            else:
+                self.comm.gather(self.wire, self.gather_buf, dst=0)
+        if self.is_master:
+            with t.phase("decode"):
+                okb = self.graph_split and self._graphB_step()
+                if not okb:
+                    if self.wc.reducible:
+                        grad = self.wire
+                    else:
+                        self.agg.zero_()
+                        self.wc.decode_all(self.gather_buf, self.agg)
+                        grad = self.agg
+                    self.opt.lr = self.lr
+                    self._apply(grad)
+        self._last_contrib = self.num_workers
+        self._post_step()
         return self.last_loss
 
     def _step_body(self) -> None:

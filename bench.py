@@ -86,7 +86,7 @@ def main():
         dedicated_ps=a.dedicated_ps,
         seed=42,
         device=device,
-        use_graph=(comm.world == 1) if a.graph is None else a.graph,
+        use_graph=True if a.graph is None else a.graph,
         overlap=(a.code == "svd") if a.overlap is None else a.overlap,
         defer_loss=True,
         amp=a.amp,

@@ -206,3 +206,33 @@ def test_gpu_whole_step_graph(code, dev):
         w1 = trainer.wire.clone()
         trainer.train_step(x, y)
         assert not torch.equal(w1, trainer.wire)
+
+
+@pytest.mark.parametrize("code", ["sgd", "svd"])
+def test_gpu_split_graphs(code, dev):
+    """Split graphs (pre-comm fwd/bwd+encode graph, post-comm decode+apply
+    graph with the collective between replays) are the N>1 path of the
+    driver's scaling run; exercised here at world=1 by flipping the mode
+    on (the un-initialized comm's gather degenerates to a row copy, which
+    is semantically identical)."""
+    from atomo_amd.data import make_loaders
+
+    trainer = _trainer(code, dev, use_graph=True)
+    # force the split mode the 8-GPU run would select
+    trainer.graph_whole = False
+    trainer._wgraph = None
+    trainer.graph_split = True
+    if trainer.gather_buf is None and not trainer.wc.reducible:
+        trainer.gather_buf = torch.zeros(
+            1, trainer.wc.total_words, device=dev
+        )
+    train, _ = make_loaders("cifar10", 64, 64, dev, seed=5)
+    it = iter(train)
+    losses = []
+    for _ in range(20):
+        x, y = next(it)
+        losses.append(trainer.train_step(x, y))
+    assert trainer.graph_split, "split-graph capture fell back to eager"
+    assert trainer._graphA is not None and trainer._graphB is not None
+    assert all(not math.isnan(l) for l in losses)
+    assert sum(losses[-5:]) < sum(losses[:5]), losses
