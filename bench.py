@@ -55,6 +55,9 @@ def main():
     p.add_argument("--wire-dtype", type=str, default="fp32",
                    choices=["fp32", "bf16"],
                    help="comm dtype for weight push + svd factor packets")
+    p.add_argument("--step-times", type=str, default=None,
+                   help="write per-step wall times (rank 0, ms, JSON) here; "
+                        "adds a device sync per step")
     a = p.parse_args()
 
     from atomo_amd.codings import make_codec
@@ -167,12 +170,35 @@ def main():
     trainer.wc.reset_device_msg_bytes()  # count timed steps only
     if a.phase_log:
         trainer.timers.sync_cuda = device.type == "cuda"
+    step_ms = [] if a.step_times else None
     t0 = time.perf_counter()
     for _ in range(a.steps):
         x, y = next(it)
-        trainer.train_step(x, y)
+        if step_ms is not None:
+            ts = time.perf_counter()
+            trainer.train_step(x, y)
+            if device.type == "cuda":
+                torch.cuda.synchronize()
+            step_ms.append(1e3 * (time.perf_counter() - ts))
+        else:
+            trainer.train_step(x, y)
     sync()
     elapsed = time.perf_counter() - t0
+    if step_ms is not None and comm.rank == 0:
+        qs = sorted(step_ms)
+        n = len(qs)
+        with open(a.step_times, "w") as f:
+            json.dump(
+                {
+                    "steps_ms": step_ms,
+                    "p5": qs[int(0.05 * n)],
+                    "p50": qs[n // 2],
+                    "p95": qs[min(n - 1, int(0.95 * n))],
+                    "p99": qs[min(n - 1, int(0.99 * n))],
+                    "max": qs[-1],
+                },
+                f,
+            )
     if a.phase_log or comm.world > 1:
         # per-rank phase breakdown (always on for N>1: the comm phase per
         # rank is the scaling diagnostic).  Without --phase-log these are

@@ -106,6 +106,9 @@ def main(argv=None):
     p.add_argument("--bucket-size", type=int, default=512)
     p.add_argument("--eval-freq", type=int, default=50)
     p.add_argument("--seed", type=int, default=42)
+    p.add_argument("--seeds", type=int, default=1,
+                   help="repeat each codec over N student seeds (seed+i) "
+                        "and record every run")
     p.add_argument("--data-seed", type=int, default=1234)
     p.add_argument("--cpu", action="store_true")
     p.add_argument("--out", default="profiles/convergence_mixture.json")
@@ -113,16 +116,23 @@ def main(argv=None):
     device = torch.device(
         "cpu" if (a.cpu or not torch.cuda.is_available()) else "cuda:0"
     )
+    import copy
+
     results = []
+    base_seed = a.seed
     for code in a.codes.split(","):
-        r = run_code(code.strip(), a, device)
-        tail = sum(r["losses"][-20:]) / min(20, len(r["losses"]))
-        head = sum(r["losses"][:20]) / min(20, len(r["losses"]))
-        print(json.dumps({"code": r["code"], "loss_first20": head,
-                          "loss_last20": tail,
-                          "final_holdout": r["holdout"][-1]
-                          if r["holdout"] else None}), flush=True)
-        results.append(r)
+        for si in range(a.seeds):
+            aa = copy.copy(a)
+            aa.seed = base_seed + si
+            r = run_code(code.strip(), aa, device)
+            r["student_seed"] = aa.seed
+            tail = sum(r["losses"][-20:]) / min(20, len(r["losses"]))
+            head = sum(r["losses"][:20]) / min(20, len(r["losses"]))
+            print(json.dumps({"code": r["code"], "seed": aa.seed,
+                              "loss_first20": head, "loss_last20": tail,
+                              "final_holdout": r["holdout"][-1]
+                              if r["holdout"] else None}), flush=True)
+            results.append(r)
     out = {
         "task": "10-class Gaussian-prototype mixture, fresh samples every "
                 "step (non-memorizable; loss/accuracy are generalization "
