@@ -239,6 +239,7 @@ class PSTrainer:
                 self.graph_whole = bool(
                     enc is not None
                     and enc.use_kernels
+                    and not enc.exact_eigh  # hipSOLVER syevd: not capturable
                     and len(enc.kernel_set) == len(self.wc.specs)
                     and self.codec.generator is None
                 )
@@ -264,6 +265,7 @@ class PSTrainer:
                 self.graph_split = bool(
                     enc is not None
                     and enc.use_kernels
+                    and not enc.exact_eigh  # hipSOLVER syevd: not capturable
                     and len(enc.kernel_set) == len(self.wc.specs)
                     and self.codec.generator is None
                 )
@@ -485,6 +487,20 @@ class PSTrainer:
             self.last_loss = float(self._loss_tensor)
         return self.last_loss
 
+    def _quarantine_graph(self, g) -> None:
+        """A partially-captured CUDAGraph aborts the process in its
+        destructor (HIPGeneratorImpl 'graph should be registered');
+        reset it if possible and keep the object alive."""
+        if g is None:
+            return
+        try:
+            g.reset()
+        except Exception:
+            pass
+        if not hasattr(self, "_dead_graphs"):
+            self._dead_graphs = []
+        self._dead_graphs.append(g)
+
     def _post_step(self) -> None:
         self.step_num += 1
         if self.watchdog is not None:
@@ -510,6 +526,7 @@ class PSTrainer:
 
     def _graphA_step(self, x: torch.Tensor, y: torch.Tensor) -> bool:
         if self._graphA is None:
+            g = None
             try:
                 self._static_x = x.clone()
                 self._static_y = y.clone()
@@ -529,6 +546,7 @@ class PSTrainer:
                     flush=True,
                 )
                 self.graph_split = False
+                self._quarantine_graph(g)
                 return False
         self._static_x.copy_(x)
         self._static_y.copy_(y)
@@ -546,6 +564,7 @@ class PSTrainer:
         """PS-side decode+apply graph (capture needs kernels warmed by an
         eager step 0; recaptures when lr changes)."""
         if self._graphB is None or self._graphB_lr != self.lr:
+            g = None
             try:
                 self.opt.lr = self.lr
                 g = torch.cuda.CUDAGraph()
@@ -563,6 +582,7 @@ class PSTrainer:
                     f"[atomo] split-graph B capture failed ({exc}); eager",
                     flush=True,
                 )
+                self._quarantine_graph(g)
                 self._graphB = None
                 self.graph_split = False
                 return False
@@ -627,6 +647,7 @@ class PSTrainer:
         standard path."""
         enc = self.wc._batched_encoder
         if self._wgraph is None or self._wgraph_lr != self.lr:
+            g = None
             try:
                 self._wgraph = None
                 self.opt.lr = self.lr
@@ -650,6 +671,7 @@ class PSTrainer:
                     flush=True,
                 )
                 self.graph_whole = False
+                self._quarantine_graph(g)
                 self._wgraph = None
                 return False
         self._static_x.copy_(x)
