@@ -341,7 +341,9 @@ class PSTrainer:
                 self.model.train()
                 if self.overlap:
                     self.wc.arm_overlap()
-                if self.use_graph:
+                if self.use_graph and not self.graph_split:
+                    # (split mode's eager step 0 must not capture the
+                    # fwd/bwd-only graph it would never replay)
                     self._fwd_bwd_graphed(x, y)
                 else:
                     self.flat_grad.zero_()
