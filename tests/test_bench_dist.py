@@ -41,5 +41,6 @@ def test_bench_under_torchrun_two_ranks():
     # stderr carries the preflight verdict and a phase line per rank
     pre = [l for l in out.stderr.splitlines() if '"preflight"' in l]
     assert pre and json.loads(pre[0])["preflight"] == "ok"
-    phases = [l for l in out.stderr.splitlines() if '"comm_s"' in l]
-    assert len(phases) == 2, out.stderr[-2000:]
+    # the two ranks share the stderr pipe, so their records can land on
+    # one line — count occurrences, not lines
+    assert out.stderr.count('"comm_s"') == 2, out.stderr[-2000:]
