@@ -231,6 +231,14 @@ class PSTrainer:
         if self.use_graph and self.amp:
             # the captured bodies run fp32; don't silently drop autocast
             self.use_graph = False
+        if (
+            self.use_graph
+            and self.is_master
+            and type(self.opt).__name__ == "ExternalAdam"
+        ):
+            # Adam's per-step bias corrections are scalar kernel args — a
+            # captured apply would freeze them at the capture step
+            self.use_graph = False
         if self.use_graph and comm.world == 1 and not self.dedicated_ps:
             if self.wc.reducible:
                 self.graph_whole = True
