@@ -211,9 +211,10 @@ class Comm:
                         self._wt.append(th)
             workers = [w for w in range(self.world) if w != dst]
             contrib = 0
-            want = target if target > 0 else (
-                len(workers) + (1 if self_counts else 0)
-            )
+            avail = len(workers) + (1 if self_counts else 0)
+            # clamp: a --num-aggregate above the contributor count would
+            # otherwise wait forever
+            want = min(target, avail) if target > 0 else avail
 
             def consume(w, rstep):
                 nonlocal contrib
