@@ -60,6 +60,12 @@ SMALL_SM = 64      # LDS-Jacobi (64-variant) + batched_gram kernel
 # sm > that goes to batched hipSOLVER syevd on device (ResNet-50/152 have
 # 1x1-conv folds up to 2048x1024 — host LAPACK costs 30-60 ms each there).
 J128_SM = int(os.environ.get("ATOMO_JACOBI_CAP", "64"))
+# randomized mode routes folds ABOVE this to the randomized solver too:
+# the LDS Jacobi is one-WG-per-matrix and its launch runtime is the
+# single biggest matrix's round count (sm=64: ~0.5 ms on one CU), while
+# a 33..64 fold amortizes into the solver's batched GEMMs + merged
+# dense eigh for ~nothing (same unbiased-projection semantics)
+JAC_SM_RAND = int(os.environ.get("ATOMO_JACOBI_SM", "32"))
 SOLVER_SM = 4096
 
 # measured per-matrix costs (ms) on MI355X + EPYC host, fp32 — used ONLY
@@ -242,9 +248,11 @@ class BatchedSVDEncoder:
             # group-level routing for big folds: batched hipSOLVER syevd is
             # nearly count-free per call, pooled host LAPACK wins for one or
             # two small-ish matrices (it overlaps the device Jacobi)
+            jac_cap = SMALL_SM if self.exact_eigh else JAC_SM_RAND
+            self._jac_cap = jac_cap
             counts = defaultdict(int)
             for i, s in enumerate(specs):
-                if self.small[i] > J128_SM:
+                if self.small[i] > (J128_SM if self.exact_eigh else jac_cap):
                     counts[self.small[i]] += 1
             solver_dims = set()
             for sm, cnt in counts.items():
@@ -271,7 +279,11 @@ class BatchedSVDEncoder:
                     s.meta["padded"] == s.numel
                     and s.meta["r_max"] <= R_CAP
                     and sm <= SOLVER_SM
-                    and (sm <= J128_SM or sm in solver_dims)
+                    and (
+                        sm <= jac_cap
+                        or (self.exact_eigh and sm <= J128_SM)
+                        or sm in solver_dims
+                    )
                 )
                 if not device_ok:
                     continue
@@ -301,11 +313,11 @@ class BatchedSVDEncoder:
                     )
                 else:
                     self.sel_mm_layers.append(i)
-                if sm <= SMALL_SM:
+                if sm <= jac_cap:
                     rows_j64.append(row)
                     for c in range((tall + GRAM_CHUNK - 1) // GRAM_CHUNK):
                         gram_work.append([row, c])
-                elif sm <= J128_SM:
+                elif self.exact_eigh and sm <= J128_SM:
                     rows_j128.append(row)
                     self.big_gram_layers.append(i)
                 else:
