@@ -307,12 +307,9 @@ class BatchedSVDEncoder:
                 sel_fit = (
                     sm * min(R_CAP, s.meta["r_max"]) * 4 + 160 <= SEL_LDS_CAP
                 )
-                if sel_fit:
-                    sel_elems = max(
-                        sel_elems, sm * min(R_CAP, s.meta["r_max"])
-                    )
-                else:
+                if not sel_fit:
                     self.sel_mm_layers.append(i)
+                bmm_sel = False
                 if sm <= jac_cap:
                     rows_j64.append(row)
                     for c in range((tall + GRAM_CHUNK - 1) // GRAM_CHUNK):
@@ -323,7 +320,13 @@ class BatchedSVDEncoder:
                 else:
                     self.solver_layers.append(i)
                     self.big_gram_layers.append(i)
-                if sel_fit:
+                    # randomized mode: selection rides the shape-grouped
+                    # rocBLAS bmm (see _sel_big_folds), not the LDS kernel
+                    bmm_sel = not self.exact_eigh
+                if sel_fit and not bmm_sel:
+                    sel_elems = max(
+                        sel_elems, sm * min(R_CAP, s.meta["r_max"])
+                    )
                     for c in range((tall + SEL_CHUNK - 1) // SEL_CHUNK):
                         sel_work.append([row, c])
             if desc_rows:
@@ -373,10 +376,18 @@ class BatchedSVDEncoder:
                     if j64_ids
                     else None
                 )
-                # big-fold Grams batched by identical fold shape: ONE
-                # stack + bmm + scatter per shape group (ResNet-152 has 97
-                # big folds in ~8 shapes) instead of a per-layer mm loop
+                # big-fold Grams AND selection batched by identical fold
+                # shape: ONE cat + bmm + scatter per shape group
+                # (ResNet-152 has 97 big folds in ~8 shapes) instead of
+                # per-layer loops or the LDS sel kernel (whose per-element
+                # inner product runs at ~4 % of HBM peak on big folds —
+                # these shapes are exactly what rocBLAS MFMA kernels eat).
+                # The stacked-A buffer is shared between the gram and sel
+                # phases within a step (_ensure_a_stacks).
                 self._gram_shape_groups = []
+                self._a_stacks = {}
+                self._a_tick = -1
+                self._enc_tick = 0
                 if not self.exact_eigh and self.big_gram_layers:
                     by_shape = defaultdict(list)
                     for i in self.big_gram_layers:
@@ -391,6 +402,42 @@ class BatchedSVDEncoder:
                                 for i in idxs
                             ]
                         ).to(dev)
+                        # selection-side tables (solver layers that fit the
+                        # stage; LDS-oversize ones keep _sel_oversize)
+                        sel_idxs = [i for i in idxs if i not in self.sel_mm_layers]
+                        sel_gather = sel_scatter = sel_rows_t = None
+                        if sel_idxs:
+                            if len(sel_idxs) != len(idxs):
+                                sel_rows_t = torch.tensor(
+                                    [idxs.index(i) for i in sel_idxs],
+                                    dtype=torch.int64,
+                                    device=dev,
+                                )
+                            r_max = specs[sel_idxs[0]].meta["r_max"]
+                            tall = max(m, n)
+                            sg, sc = [], []
+                            base_sel = torch.arange(
+                                sm * r_max, dtype=torch.int64
+                            )
+                            base_out = torch.arange(
+                                r_max * tall, dtype=torch.int64
+                            )
+                            for i in sel_idxs:
+                                so = self.stage_offsets[i]
+                                wo = specs[i].wire_offset
+                                sg.append(
+                                    base_sel + (so + 1 + r_max * (1 + sm))
+                                )
+                                # bmm output (r_max, tall) row-major goes to
+                                # the u region (tall fold) or v region (wide)
+                                out_off = (
+                                    wo + 1
+                                    if m >= n
+                                    else wo + 1 + r_max * (m + 1)
+                                )
+                                sc.append(base_out + out_off)
+                            sel_gather = torch.cat(sg).to(dev)
+                            sel_scatter = torch.cat(sc).to(dev)
                         self._gram_shape_groups.append(
                             (
                                 m,
@@ -398,8 +445,59 @@ class BatchedSVDEncoder:
                                 m >= n,
                                 [param_offsets[i] for i in idxs],
                                 scat,
+                                sel_idxs,
+                                sel_gather,
+                                sel_scatter,
+                                sel_rows_t,
                             )
                         )
+                    # one global stage->wire copy for the non-GEMM packet
+                    # parts of every shape-grouped layer: header, s_wire,
+                    # and the small factor (facT == v^T for tall folds,
+                    # u^T for wide — identical row-major layout)
+                    src, dst = [], []
+                    for (m, n), idxs in sorted(by_shape.items()):
+                        sm = min(m, n)
+                        for i in idxs:
+                            if i in self.sel_mm_layers:
+                                continue
+                            r_max = specs[i].meta["r_max"]
+                            so = self.stage_offsets[i]
+                            wo = specs[i].wire_offset
+                            # header
+                            src.append(torch.tensor([so], dtype=torch.int64))
+                            dst.append(torch.tensor([wo], dtype=torch.int64))
+                            # s_wire
+                            src.append(
+                                torch.arange(r_max, dtype=torch.int64)
+                                + (so + 1)
+                            )
+                            dst.append(
+                                torch.arange(r_max, dtype=torch.int64)
+                                + (wo + 1 + r_max * m)
+                            )
+                            # small factor (r_max, sm) row-major
+                            fac_dst = (
+                                wo + 1 + r_max * (m + 1)
+                                if m >= n
+                                else wo + 1
+                            )
+                            src.append(
+                                torch.arange(r_max * sm, dtype=torch.int64)
+                                + (so + 1 + r_max)
+                            )
+                            dst.append(
+                                torch.arange(r_max * sm, dtype=torch.int64)
+                                + fac_dst
+                            )
+                    self._sel_copy_src = (
+                        torch.cat(src).to(dev) if src else None
+                    )
+                    self._sel_copy_dst = (
+                        torch.cat(dst).to(dev) if dst else None
+                    )
+                else:
+                    self._sel_copy_src = self._sel_copy_dst = None
                 self.eval_offs_dev = torch.tensor(eval_offs, dtype=torch.int64, device=dev)
                 self.eval_offs = eval_offs
                 self.evals_dev = torch.zeros(max(1, ev_off), dtype=torch.float32, device=dev)
@@ -637,18 +735,77 @@ class BatchedSVDEncoder:
         ) % (1 << 62)
         self.seed_host[0] = self._seed
 
+    def _ensure_a_stacks(self, flat_grad: torch.Tensor) -> None:
+        """Copy every shape group's folds into its preallocated stacked
+        (B, m, n) buffer once per encode (shared by the gram bmm and the
+        selection bmm; cat(out=) keeps storage stable for graph capture)."""
+        if self._a_tick == self._enc_tick:
+            return
+        for grp in self._gram_shape_groups:
+            m, n, tall, offs = grp[0], grp[1], grp[2], grp[3]
+            key = (m, n)
+            buf = self._a_stacks.get(key)
+            if buf is None:
+                buf = torch.empty(
+                    len(offs) * m * n, dtype=torch.float32, device=self.device
+                )
+                self._a_stacks[key] = buf
+            torch.cat(
+                [flat_grad.narrow(0, o, m * n) for o in offs], out=buf
+            )
+        self._a_tick = self._enc_tick
+
     def _compute_big_grams(self, flat_grad: torch.Tensor) -> None:
-        """Batched big-fold Grams: one stack + bmm + scatter per identical
+        """Batched big-fold Grams: one cat + bmm + scatter per identical
         fold shape (randomized mode; replaces the per-layer rocBLAS loop)."""
-        for m, n, tall, offs, scat in self._gram_shape_groups:
-            a = torch.stack(
-                [flat_grad.narrow(0, o, m * n) for o in offs]
-            ).view(len(offs), m, n)
+        self._ensure_a_stacks(flat_grad)
+        for grp in self._gram_shape_groups:
+            m, n, tall, offs, scat = grp[0], grp[1], grp[2], grp[3], grp[4]
+            a = self._a_stacks[(m, n)].view(len(offs), m, n)
             if tall:
                 g = torch.bmm(a.transpose(1, 2), a)
             else:
                 g = torch.bmm(a, a.transpose(1, 2))
             self.grams.index_copy_(0, scat, g.reshape(-1))
+
+    def _sel_big_folds(self, flat_grad: torch.Tensor, wire: torch.Tensor):
+        """Selection GEMMs for solver layers, batched per fold shape:
+        long factor = bmm(sel^T, A^T or A) straight into the wire via a
+        precomputed scatter; header + s_wire + small factor ship in ONE
+        global stage->wire gather/copy.  Replaces the LDS batched_sel for
+        big folds (whose per-element inner loop ran at ~4 % of HBM peak);
+        rows past the sampled r_hat are zeros and decode ignores them."""
+        if not self._gram_shape_groups:
+            return
+        self._ensure_a_stacks(flat_grad)
+        if self._sel_copy_src is not None:
+            wire.index_copy_(
+                0,
+                self._sel_copy_dst,
+                self.stage_dev.index_select(0, self._sel_copy_src),
+            )
+        for grp in self._gram_shape_groups:
+            (m, n, tall, offs, _, sel_idxs, sel_gather, sel_scatter,
+             sel_rows_t) = grp
+            if not sel_idxs:
+                continue
+            B = len(sel_idxs)
+            sm = min(m, n)
+            r_max = self.specs[sel_idxs[0]].meta["r_max"]
+            sel = (
+                self.stage_dev.index_select(0, sel_gather)
+                .view(B, sm, r_max)
+            )
+            a = self._a_stacks[(m, n)].view(len(offs), m, n)
+            if sel_rows_t is not None:
+                # group mixes bmm-sel and LDS-oversize layers (cannot
+                # happen for homogeneous shapes, kept for safety)
+                a = a.index_select(0, sel_rows_t)
+            if tall:
+                out = torch.bmm(sel.transpose(1, 2), a.transpose(1, 2))
+            else:
+                out = torch.bmm(sel.transpose(1, 2), a)
+            wire.index_copy_(0, sel_scatter, out.reshape(-1))
 
     def _dense_eigh(self, s: torch.Tensor, sweeps: int = 6):
         """Batched symmetric eigh via the one-wave-per-matrix LDS Jacobi
@@ -913,6 +1070,8 @@ class BatchedSVDEncoder:
         grams_done: bool = False,
     ) -> int:
         use_kernels = self.use_kernels and flat_grad is not None
+        if use_kernels:
+            self._enc_tick += 1  # invalidates the shared stacked-A buffers
         kernel_set = self.kernel_set if use_kernels else set()
         specs = self.specs
         marks = [time.perf_counter()] if _TRACE else None
@@ -1072,6 +1231,7 @@ class BatchedSVDEncoder:
                     flat_grad, wire, self.stage_dev, self.desc, self.sel_work,
                     self.sel_work.shape[0], self.sel_elems,
                 )
+            self._sel_big_folds(flat_grad, wire)
             self._sel_oversize(flat_grad, wire)
             if not host_layers:
                 mark("async sample+sel")
@@ -1206,6 +1366,7 @@ class BatchedSVDEncoder:
                     flat_grad, wire, self.stage_dev, self.desc, self.sel_work,
                     self.sel_work.shape[0], self.sel_elems,
                 )
+            self._sel_big_folds(flat_grad, wire)
             self._sel_oversize(flat_grad, wire)
         sd = self.stage_dev
         for i, r_hat in host_plans:
