@@ -1,32 +1,39 @@
-"""Batched SVD wire encoder — at most one host sync per step, zero in the
-fully-device configuration.
+"""Batched SVD wire encoder — zero host syncs per step in the fully-device
+configuration, and whole-step hipGraph-capturable.
 
 The naive per-layer path costs ~60 host round trips per step (factorize,
 sample, write).  This encoder restructures the work MI355X-first:
 
 DEVICE layers (even 2-D fold, wire budget r_max <= 32):
-  * small-dim <= 64: batched_gram kernel -> warm-started LDS Jacobi
-    (jacobi_eigh_kernel, previous step's eigenbasis pre-rotates so sweeps
-    converge in ~1-2 iterations).
-  * bigger folds (1x1-conv Grams up to 2048x1024): rocBLAS Gram ->
-    warm-started randomized Rayleigh-Ritz (_solve_big_folds_randomized:
-    one subspace-iteration power step + CholQR2 + ONE merged tiny eigh
-    across every size group — the north-star "one-pass randomized SVD";
-    the exact batched hipSOLVER syevd survives as the ATOMO_EXACT_EIGH=1
-    oracle, size-groups merged by zero-padding per the measured cost
-    model).
+  * small-dim <= 32 (ATOMO_JACOBI_SM): batched_gram kernel -> warm-started
+    LDS Jacobi (jacobi_eigh_kernel; the previous step's eigenbasis
+    pre-rotates so 2 capped sweeps suffice; launched on a fork/join side
+    stream so its one-CU-per-matrix critical path overlaps the chip-wide
+    solver GEMMs).
+  * bigger folds (1x1-conv Grams up to 2048x1024, AlexNet-227 fc folds):
+    shape-grouped cat+bmm Grams -> warm-started randomized Rayleigh-Ritz
+    (_solve_big_folds_randomized: one subspace-iteration power step +
+    column-scaled Lowdin orthonormalization x2 + ONE merged tiny eigh on
+    the one-wave-per-matrix jacobi_dense kernel across every size group —
+    the north-star "one-pass randomized SVD"; the wire ships u = A v / s,
+    so decoded atoms are A v v^T / p: unbiased for the gradient's
+    projection onto the computed subspace for ANY orthonormal V, with the
+    only bias being the tracked tail energy).  The exact batched
+    hipSOLVER syevd survives as the ATOMO_EXACT_EIGH=1 oracle.
   * then: fused on-device Bernoulli sampler + stage builder
-    (sample_stage_kernel; counter-hash RNG; Msg bytes counted on device)
-    -> batched_sel kernel writes the wire packets.  No host sync at all
-    when no host layers exist and no host RNG generator is pinned.
+    (sample_stage_kernel; counter-hash RNG read from a device seed buffer
+    so hipGraph replays draw fresh atoms; Msg bytes counted on device)
+    -> wire packets via the LDS batched_sel kernel (small folds), shape-
+    grouped rocBLAS bmms (big folds, stacked-A shared with the gram
+    phase), or per-layer rocBLAS at fixed r_max (LDS-oversize folds).
 HOST layers (odd zero-padded 1-D folds, r_max > 32, or sm > 4096):
   rocBLAS Gram, robust LAPACK eigh on a warmed thread pool (overlapping
   the device work), host sampling, staged factors H2D, rocBLAS selection
   GEMMs straight into the wire.
-OVERLAP mode (--overlap): per-layer Gram hooks fire during backward; each
-  solver call-group's eigh and each host layer's LAPACK solve launch from a
-  pool thread the moment their last Gram lands — deep layers' gradients
-  materialize first, so most eigensolve time hides under backward.
+OVERLAP mode (--overlap): per-layer Gram hooks fire during backward for
+  the layers whose post-gram work benefits (host-LAPACK layers; big folds
+  in exact-oracle mode, whose solver groups launch from pool threads);
+  everything else runs post-backward in a handful of batched launches.
 
 Semantics identical to SVDCodec.encode_into (same wire layout, same
 sampler, same unbiasedness invariant E[sum s_i/p_i u_i v_i^T] = grad;
