@@ -44,3 +44,44 @@ def test_bench_under_torchrun_two_ranks():
     # the two ranks share the stderr pipe, so their records can land on
     # one line — count occurrences, not lines
     assert out.stderr.count('"comm_s"') == 2, out.stderr[-2000:]
+
+
+def test_multinode_rendezvous_two_agents():
+    """Multi-node launch path (tools/run_node.sh shape): TWO torchrun
+    agents (--nnodes=2, node ranks 0/1) rendezvous on localhost and train
+    — upgrades the cluster-tooling story from documented-only to
+    exercised (reference: tools/pytorch_ec2.py cluster launch)."""
+    import time
+
+    agents = []
+    for node_rank in (0, 1):
+        cmd = [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=2", "--node-rank", str(node_rank),
+            "--nproc-per-node", "1",
+            "--master-addr", "127.0.0.1", "--master-port", "29671",
+            "bench.py", "--cpu", "--network", "LeNet", "--dataset", "mnist",
+            "--steps", "3", "--warmup", "1", "--batch-size", "16",
+        ]
+        env = dict(os.environ)
+        env["OMP_NUM_THREADS"] = "2"
+        agents.append(
+            subprocess.Popen(cmd, cwd=REPO, env=env,
+                             stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+                             text=True)
+        )
+    outs = []
+    deadline = time.time() + 300
+    for p in agents:
+        out, err = p.communicate(timeout=max(10, deadline - time.time()))
+        outs.append((p.returncode, out, err))
+    assert all(rc == 0 for rc, _, _ in outs), [
+        (rc, err[-800:]) for rc, _, err in outs
+    ]
+    # node 0 hosts global rank 0 -> the single result line
+    results = [
+        json.loads(line)
+        for line in outs[0][1].splitlines()
+        if line.startswith("{") and '"metric"' in line
+    ]
+    assert len(results) == 1 and results[0]["n_gpus"] == 2
