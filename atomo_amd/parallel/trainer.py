@@ -547,7 +547,7 @@ class PSTrainer:
                         self._body_A()
                 torch.cuda.current_stream().wait_stream(side)
                 g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
+                with torch.cuda.graph(g, capture_error_mode="thread_local"):
                     self._body_A()
                 self._graphA = g
             except Exception as exc:
@@ -578,7 +578,7 @@ class PSTrainer:
             try:
                 self.opt.lr = self.lr
                 g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
+                with torch.cuda.graph(g, capture_error_mode="thread_local"):
                     if self.wc.reducible:
                         self._apply(self.flat_grad)
                     else:
@@ -670,7 +670,9 @@ class PSTrainer:
                         self._step_body()
                 torch.cuda.current_stream().wait_stream(side)
                 g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
+                # thread_local: NCCL watchdog threads issue event queries
+                # that would fail a global-mode capture at world > 1
+                with torch.cuda.graph(g, capture_error_mode="thread_local"):
                     self._step_body()
                 self._wgraph = g
                 self._wgraph_lr = self.lr
@@ -711,7 +713,7 @@ class PSTrainer:
                         loss.backward()
                 torch.cuda.current_stream().wait_stream(side)
                 g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
+                with torch.cuda.graph(g, capture_error_mode="thread_local"):
                     self.flat_grad.zero_()
                     self._static_loss = self.loss_fn(
                         self.model(self._static_x), self._static_y
