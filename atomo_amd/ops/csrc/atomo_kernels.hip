@@ -187,7 +187,11 @@ __global__ void qsgd_unpack_acc_kernel(const float* __restrict__ norms,
 __global__ void qsgd_pack_batched_kernel(
     const float* __restrict__ flat, float* __restrict__ wire,
     const int64_t* __restrict__ desc, const int32_t* __restrict__ work,
-    int n_tiles, int bucket_size, int qlevel, bool terngrad, uint64_t seed) {
+    int n_tiles, int bucket_size, int qlevel, bool terngrad,
+    const unsigned long long* __restrict__ seed_buf) {
+  // seed from device memory: a captured H2D copy refreshes it per graph
+  // replay so every replay rolls fresh stochastic-rounding dice
+  const uint64_t seed = seed_buf[0];
   extern __shared__ uint8_t lds_codes[];
   const int waves_per_wg = blockDim.x / WAVE;
   const int wave_id = threadIdx.x / WAVE;
@@ -495,7 +499,8 @@ void atomo_svd_decode_acc_launch(const float* regions, float* out, int W,
 void atomo_qsgd_pack_batched_launch(const float* flat, float* wire,
                                     const int64_t* desc, const int32_t* work,
                                     int n_tiles, int bucket_size, int qlevel,
-                                    bool terngrad, uint64_t seed,
+                                    bool terngrad,
+                                    const unsigned long long* seed_buf,
                                     hipStream_t stream) {
   const int block = 256;
   const int waves_per_wg = block / WAVE;
@@ -503,7 +508,7 @@ void atomo_qsgd_pack_batched_launch(const float* flat, float* wire,
   const int grid = grid_for((int64_t)n_tiles * WAVE, block);
   hipLaunchKernelGGL(qsgd_pack_batched_kernel, dim3(grid), dim3(block), lds,
                      stream, flat, wire, desc, work, n_tiles, bucket_size,
-                     qlevel, terngrad, seed);
+                     qlevel, terngrad, seed_buf);
 }
 
 void atomo_qsgd_unpack_batched_launch(const float* wire, float* agg,

@@ -28,7 +28,7 @@ void atomo_svd_decode_batched_launch(const float*, int64_t, int, float*,
                                      hipStream_t);
 void atomo_qsgd_pack_batched_launch(const float*, float*, const int64_t*,
                                     const int32_t*, int, int, int, bool,
-                                    uint64_t, hipStream_t);
+                                    const unsigned long long*, hipStream_t);
 void atomo_qsgd_unpack_batched_launch(const float*, float*, const int64_t*,
                                       const int32_t*, int, int, int,
                                       hipStream_t);
@@ -209,15 +209,20 @@ void svd_decode_batched(torch::Tensor stacked, torch::Tensor agg,
 void qsgd_pack_batched(torch::Tensor flat, torch::Tensor wire,
                        torch::Tensor desc, torch::Tensor work,
                        int64_t n_tiles, int64_t bucket_size, int64_t qlevel,
-                       bool terngrad, int64_t seed) {
+                       bool terngrad, torch::Tensor seed_dev) {
   check_f32_cuda(flat, "flat");
   check_f32_cuda(wire, "wire");
+  TORCH_CHECK(seed_dev.is_cuda() && seed_dev.scalar_type() == torch::kInt64,
+              "seed_dev must be cuda int64");
   TORCH_CHECK(bucket_size <= 8192, "bucket_size too large for LDS staging");
   if (n_tiles == 0) return;
   atomo_qsgd_pack_batched_launch(
       flat.data_ptr<float>(), wire.data_ptr<float>(),
       desc.data_ptr<int64_t>(), work.data_ptr<int32_t>(), (int)n_tiles,
-      (int)bucket_size, (int)qlevel, terngrad, (uint64_t)seed, cur_stream());
+      (int)bucket_size, (int)qlevel, terngrad,
+      reinterpret_cast<const unsigned long long*>(
+          seed_dev.data_ptr<int64_t>()),
+      cur_stream());
 }
 
 void qsgd_unpack_batched(torch::Tensor wire, torch::Tensor agg,

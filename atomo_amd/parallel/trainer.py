@@ -242,6 +242,8 @@ class PSTrainer:
         if self.use_graph and comm.world == 1 and not self.dedicated_ps:
             if self.wc.reducible:
                 self.graph_whole = True
+            elif self.wc._qsgd_tables is not None:
+                self.graph_whole = True  # batched pack/unpack, device seed
             else:
                 enc = self.wc._batched_encoder
                 self.graph_whole = bool(
@@ -267,6 +269,8 @@ class PSTrainer:
             and not self.wire_bf16
         ):
             if self.wc.reducible:
+                self.graph_split = True
+            elif self.wc._qsgd_tables is not None:
                 self.graph_split = True
             else:
                 enc = self.wc._batched_encoder
@@ -560,9 +564,7 @@ class PSTrainer:
                 return False
         self._static_x.copy_(x)
         self._static_y.copy_(y)
-        enc = self.wc._batched_encoder
-        if enc is not None:
-            enc.advance_seed()
+        self.wc.advance_seeds()
         self._graphA.replay()
         if self.defer_loss:
             self._loss_tensor = self._static_loss
@@ -655,7 +657,6 @@ class PSTrainer:
         """Capture-once/replay whole-step graph.  Returns False (and
         disables itself) if capture fails — caller falls back to the
         standard path."""
-        enc = self.wc._batched_encoder
         if self._wgraph is None or self._wgraph_lr != self.lr:
             g = None
             try:
@@ -688,8 +689,7 @@ class PSTrainer:
                 return False
         self._static_x.copy_(x)
         self._static_y.copy_(y)
-        if enc is not None:
-            enc.advance_seed()  # captured H2D copy re-reads the pinned seed
+        self.wc.advance_seeds()  # captured H2D copy re-reads pinned seeds
         self._wgraph.replay()
         if self.defer_loss:
             self._loss_tensor = self._static_loss

@@ -91,6 +91,14 @@ class WireCodec:
                     ),
                 }
                 self._qsgd_seed = (987654321 ^ rank_mix) % (1 << 62)
+                # device-resident sampler seed (refreshed via pinned host
+                # memory) so hipGraph replays roll fresh rounding dice
+                self._qsgd_seed_host = torch.zeros(
+                    1, dtype=torch.int64, pin_memory=True
+                )
+                self._qsgd_seed_dev = torch.zeros(
+                    1, dtype=torch.int64, device=device
+                )
         # scratch for layers whose 2-D fold is zero-padded (odd 1-D sizes)
         self._pad_scratch = {}
         if isinstance(codec, SVDCodec) and codec.compress:
@@ -247,6 +255,19 @@ class WireCodec:
             h.remove()
         self._overlap_handles = []
 
+    def advance_seeds(self) -> None:
+        """Host-side LCG step for whichever device sampler this codec
+        uses; under a whole-step/split hipGraph the trainer calls this
+        before each replay (the captured pinned-memory copy re-reads the
+        scalar at replay time)."""
+        if self._batched_encoder is not None:
+            self._batched_encoder.advance_seed()
+        if self._qsgd_tables is not None:
+            self._qsgd_seed = (
+                self._qsgd_seed * 6364136223846793005 + 1442695040888963407
+            ) % (1 << 62)
+            self._qsgd_seed_host[0] = self._qsgd_seed
+
     def arm_overlap(self) -> None:
         if self._batched_encoder is not None:
             self._batched_encoder.arm_overlap()
@@ -265,16 +286,23 @@ class WireCodec:
         if overlap_done and isinstance(self.codec, QSGDCodec):
             return sum(s.wire_words for s in self.specs)
         if self._qsgd_tables is not None and flat_grad is not None:
+            import torch as _t
+
             from ..ops import ext
 
             t = self._qsgd_tables
-            self._qsgd_seed = (
-                self._qsgd_seed * 6364136223846793005 + 1442695040888963407
-            ) % (1 << 62)
+            if _t.cuda.is_current_stream_capturing():
+                # captured H2D copy re-reads the pinned scalar per replay
+                self._qsgd_seed_dev.copy_(
+                    self._qsgd_seed_host, non_blocking=True
+                )
+            else:
+                self.advance_seeds()
+                self._qsgd_seed_dev.fill_(self._qsgd_seed)
             ext().qsgd_pack_batched(
                 flat_grad, wire, t["desc"], t["pack"], t["pack"].shape[0],
                 self.codec.bucket_size, self.codec.qlevel,
-                self.codec.scheme == "terngrad", self._qsgd_seed,
+                self.codec.scheme == "terngrad", self._qsgd_seed_dev,
             )
             return sum(s.wire_words for s in self.specs)
         grads = [

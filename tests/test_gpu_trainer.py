@@ -173,7 +173,7 @@ def test_gpu_svd_wire_unbiased(dev):
     assert rel < 0.35, rel
 
 
-@pytest.mark.parametrize("code", ["sgd", "svd"])
+@pytest.mark.parametrize("code", ["sgd", "svd", "qsgd"])
 def test_gpu_whole_step_graph(code, dev):
     """Whole-step hipGraph (fwd/bwd+encode+decode+apply in one replay):
     must actually capture (graph_whole stays True), train, and draw FRESH
@@ -199,8 +199,9 @@ def test_gpu_whole_step_graph(code, dev):
         # sizes, so the counter must exceed a fixed-draw multiple check
         total = int(enc.used_words_dev.item())
         assert total > 0
+    if code in ("svd", "qsgd"):
         # two consecutive replays on identical data must differ in the
-        # wire (different Bernoulli draws -> different atoms/packets)
+        # wire (fresh Bernoulli / stochastic-rounding draws per replay)
         x, y = next(it)
         trainer.train_step(x, y)
         w1 = trainer.wire.clone()
