@@ -96,7 +96,12 @@ class Comm:
         (sync_replicas_master_nn.py:198-215) mapped onto isend/irecv:
         the PS calls ``on_arrival(rank)`` as each worker's bucket lands
         (its own row first), so decode overlaps the remaining receives.
-        Returns the arrival order on dst, else None."""
+        Returns the arrival order on dst, else None.
+
+        NOTE: the trainer's P2P mode now uses ``gather_partial`` (whose
+        waiter threads give true arrival order under gloo too, where
+        ``is_completed()`` never resolves); this simpler single-step
+        variant remains for callers without pipelined buffers."""
         if not self._initialized:
             if out_stacked is not None:
                 out_stacked[0].copy_(send)

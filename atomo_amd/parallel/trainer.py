@@ -141,7 +141,7 @@ class PSTrainer:
                 dtype=torch.float32,
                 device=self.device,
             )
-            if self.p2p and num_aggregate and self.is_master
+            if self.p2p and self.is_master
             else None
         )
         self.agg = (
@@ -389,65 +389,42 @@ class PSTrainer:
 
         if self.p2p and not self.wc.reducible:
             with t.phase("comm"):
-                if self.num_aggregate > 0:
-                    # TRUE partial aggregation: the PS returns after the
-                    # first K contributions; a straggler's packet is left
-                    # pending and dropped (stale) when it lands next step.
-                    # The colocated PS's own contribution is always in the
-                    # first-K set — that is physical (zero latency), the
-                    # same property the reference's waitany drain would
-                    # have; which K-1 workers fill the rest is genuine
-                    # arrival order, not rank order (ADVICE r1 fairness
-                    # note: with a persistent straggler the EXCLUDED set
-                    # is whoever is actually late, not a fixed rank)
-                    contrib = [0]
-                    if self.is_master:
-                        self.agg.zero_()
+                # Arrival-order P2P gather (the reference's waitany drain)
+                # with decode-as-arrives.  num_aggregate > 0 is TRUE
+                # partial aggregation: the PS returns after the first K
+                # contributions; a straggler's packet stays pending and
+                # is dropped as stale when it lands.  The colocated PS's
+                # own contribution is always in the first-K set — that is
+                # physical (zero latency); which K-1 workers fill the
+                # rest is genuine arrival order, not rank order (ADVICE
+                # r1 fairness note: with a persistent straggler the
+                # EXCLUDED set is whoever is actually late, not a fixed
+                # rank).
+                contrib = [0]
+                if self.is_master:
+                    self.agg.zero_()
 
-                        def on_arr(w, ph):
-                            self.wc.decode_all(
-                                self.pp_buf[ph, w : w + 1], self.agg
-                            )
-                            contrib[0] += 1
+                    def on_arr(w, sl):
+                        self.wc.decode_all(
+                            self.pp_buf[sl, w : w + 1], self.agg
+                        )
+                        contrib[0] += 1
 
-                        contrib[0] = 0
-                        self.comm.gather_partial(
-                            self.wire,
-                            self.pp_buf,
-                            self.step_num,
-                            dst=0,
-                            target=self.num_aggregate,
-                            on_arrival=on_arr,
-                            self_counts=not self.dedicated_ps,
-                        )
-                    else:
-                        self.comm.gather_partial(
-                            self.wire, None, self.step_num, dst=0,
-                            target=self.num_aggregate,
-                        )
-                        contrib[0] = self.num_aggregate
+                    self.comm.gather_partial(
+                        self.wire,
+                        self.pp_buf,
+                        self.step_num,
+                        dst=0,
+                        target=self.num_aggregate,
+                        on_arrival=on_arr,
+                        self_counts=not self.dedicated_ps,
+                    )
                 else:
-                    # full-sync arrival-order gather: the PS decodes each
-                    # worker's bucket as it lands, overlapping decode with
-                    # the remaining receives
-                    contrib = [0]
-                    if self.is_master:
-                        self.agg.zero_()
-
-                        def on_arrival(w):
-                            if self.dedicated_ps and w == 0:
-                                return
-                            self.wc.decode_all(
-                                self.gather_buf[w : w + 1], self.agg
-                            )
-                            contrib[0] += 1
-
-                        self.comm.gather_arrival(
-                            self.wire, self.gather_buf, dst=0,
-                            on_arrival=on_arrival,
-                        )
-                    else:
-                        self.comm.gather_arrival(self.wire, None, dst=0)
+                    self.comm.gather_partial(
+                        self.wire, None, self.step_num, dst=0,
+                        target=self.num_aggregate,
+                    )
+                    contrib[0] = self.num_aggregate or self.num_workers
             self._last_contrib = contrib[0]
             grad_flat = self.agg if self.is_master else None
         else:

@@ -12,11 +12,16 @@ import sys
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+def _port(base):
+    """pid-offset port: avoids TIME_WAIT collisions across test runs."""
+    return base + (os.getpid() % 400)
+
+
 def test_bench_under_torchrun_two_ranks():
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
         "--nnodes=1", "--nproc-per-node", "2",
-        "--master-addr", "127.0.0.1", "--master-port", "29651",
+        "--master-addr", "127.0.0.1", "--master-port", str(_port(29651)),
         "bench.py", "--cpu", "--network", "LeNet", "--dataset", "mnist",
         "--gpus", "2", "--steps", "3", "--warmup", "1", "--batch-size", "16",
     ]
@@ -59,7 +64,7 @@ def test_multinode_rendezvous_two_agents():
             sys.executable, "-m", "torch.distributed.run",
             "--nnodes=2", "--node-rank", str(node_rank),
             "--nproc-per-node", "1",
-            "--master-addr", "127.0.0.1", "--master-port", "29671",
+            "--master-addr", "127.0.0.1", "--master-port", str(_port(29671)),
             "bench.py", "--cpu", "--network", "LeNet", "--dataset", "mnist",
             "--steps", "3", "--warmup", "1", "--batch-size", "16",
         ]

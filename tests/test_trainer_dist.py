@@ -11,6 +11,11 @@ import pytest
 import torch
 
 
+def _port(base):
+    """pid-offset port: avoids TIME_WAIT collisions across test runs."""
+    return base + (os.getpid() % 400)
+
+
 def _run_rank(rank, world, port, code, dedicated, q):
     os.environ["RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world)
@@ -50,6 +55,7 @@ def _run_rank(rank, world, port, code, dedicated, q):
 
 
 def _launch(code, dedicated, port):
+    port = _port(port)
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     world = 2
@@ -130,6 +136,7 @@ def test_dist_p2p_mode(num_agg, port):
     aggregation both train and keep ranks weight-synchronized."""
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
+    port = _port(port)
     world = 3
     procs = [
         ctx.Process(target=_run_rank_p2p, args=(r, world, port, num_agg, q))
@@ -197,7 +204,7 @@ def test_dist_partial_aggregation_drops_straggler():
     is dropped as stale."""
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    world, port = 3, 29631
+    world, port = 3, _port(29631)
     procs = [
         ctx.Process(target=_run_rank_straggler, args=(r, world, port, q))
         for r in range(world)
@@ -249,7 +256,7 @@ def test_dist_bf16_wire():
     weight-synchronized."""
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_run_rank_bf16, args=(r, 2, 29641, q))
+    procs = [ctx.Process(target=_run_rank_bf16, args=(r, 2, _port(29641), q))
              for r in range(2)]
     for p in procs:
         p.start()
