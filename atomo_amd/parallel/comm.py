@@ -341,6 +341,9 @@ class Comm:
         self._pg_outstanding = 0
         for wq in getattr(self, "_wt_q", {}).values():
             wq.put(None)  # stop waiter threads
+        for th in getattr(self, "_wt", []):
+            th.join(timeout=10)  # join before gloo teardown (exit race)
+        self._wt = []
         self._wt_q = {}
         if hasattr(self, "_pg_recv"):
             del self._pg_recv  # force re-init on next partial use
