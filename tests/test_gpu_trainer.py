@@ -237,3 +237,41 @@ def test_gpu_split_graphs(code, dev):
     assert trainer._graphA is not None and trainer._graphB is not None
     assert all(not math.isnan(l) for l in losses)
     assert sum(losses[-5:]) < sum(losses[:5]), losses
+
+
+def test_gpu_graph_capture_with_live_nccl_group(dev):
+    """De-risk the 8-GPU scaling run's capture path: initialize a real
+    (world-1) NCCL/RCCL process group — which spawns the ProcessGroupNCCL
+    watchdog threads whose event queries fail a global-mode capture — run
+    a collective to force communicator init, then verify the whole-step
+    graph still captures and trains (capture_error_mode=thread_local)."""
+    import os
+
+    import torch.distributed as dist
+
+    from atomo_amd.data import make_loaders
+
+    created = False
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29681")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+        created = True
+    try:
+        probe = torch.ones(8, device=dev)
+        dist.all_reduce(probe)  # force communicator + watchdog init
+        trainer = _trainer("svd", dev, use_graph=True)
+        assert trainer.graph_whole
+        train, _ = make_loaders("cifar10", 64, 64, dev, seed=9)
+        it = iter(train)
+        losses = []
+        for _ in range(10):
+            x, y = next(it)
+            losses.append(trainer.train_step(x, y))
+        assert trainer.graph_whole and trainer._wgraph is not None, (
+            "capture fell back with a live NCCL group"
+        )
+        assert all(not math.isnan(l) for l in losses)
+    finally:
+        if created:
+            dist.destroy_process_group()
