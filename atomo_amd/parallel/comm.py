@@ -328,6 +328,33 @@ class Comm:
         if self._initialized:
             dist.recv(flat, src=src)
 
+    # -- straggler self-skip (the reference's tag-77 kill, redesigned) ---
+    # The reference's master sends kill signals that workers Iprobe
+    # mid-backward (resnet_split.py:458-571, unwired).  Here the PS
+    # publishes its step on the rendezvous store; a worker that has
+    # fallen >= D-2 steps behind SKIPS its forward/backward and ships a
+    # valid zero packet instead (decodes to a zero contribution the PS
+    # drops as stale anyway) — so a persistently slow rank catches up
+    # instead of throttling the depth-D pipeline, while every posted
+    # send/receive stays 1:1 matched.
+    def publish_step(self, step: int) -> None:
+        if self._initialized:
+            if not hasattr(self, "_store"):
+                self._store = dist.distributed_c10d._get_default_store()
+            self._store.set("atomo_ps_step", str(step))
+
+    def ps_step_behind(self, my_step: int) -> int:
+        """How many steps the PS is ahead of ``my_step`` (0 if unknown)."""
+        if not self._initialized:
+            return 0
+        if not hasattr(self, "_store"):
+            self._store = dist.distributed_c10d._get_default_store()
+        try:
+            ps = int(self._store.get("atomo_ps_step"))
+        except Exception:
+            return 0
+        return max(0, ps - my_step)
+
     def drain_partial(self) -> None:
         """Complete everything left pending by the partial-mode pipelines
         (safe by 1:1 matching; call before destroying the group)."""

@@ -302,6 +302,9 @@ class PSTrainer:
                 self._side_stream, self.params, self.wire
             )
 
+        self.skip_count = 0  # straggler self-skips (partial mode, worker)
+        if self.p2p and self.num_aggregate > 0 and self.is_master:
+            self.comm.publish_step(0)
         # make every rank start from rank-0's init
         self.comm.broadcast(self.flat, src=0)
 
@@ -348,7 +351,18 @@ class PSTrainer:
             else:
                 self.comm.broadcast(self.flat, src=0)
 
-        if self.is_worker:
+        skip_compute = False
+        if self.p2p and self.num_aggregate > 0 and not self.is_master:
+            # straggler self-skip (tag-77 kill redesigned, see Comm): a
+            # worker >= D-2 steps behind ships a valid zero packet and
+            # catches up instead of throttling the pipeline
+            behind = self.comm.ps_step_behind(self.step_num)
+            if behind >= self.comm.PIPE_DEPTH - 2:
+                skip_compute = True
+                self.skip_count += 1
+                self.wire.zero_()
+
+        if self.is_worker and not skip_compute:
             with t.phase("comp"):
                 self.model.train()
                 if self.overlap:
@@ -384,7 +398,7 @@ class PSTrainer:
                     )
                 if used >= 0:
                     t.add_scalar("msg_bytes", 4.0 * used)
-        elif self.wc.reducible:
+        elif self.wc.reducible and not self.is_worker:
             self.flat_grad.zero_()  # dedicated PS contributes zeros to the sum
 
         if self.p2p and not self.wc.reducible:
@@ -425,6 +439,8 @@ class PSTrainer:
                         target=self.num_aggregate,
                     )
                     contrib[0] = self.num_aggregate or self.num_workers
+            if self.num_aggregate > 0 and self.is_master:
+                self.comm.publish_step(self.step_num + 1)
             self._last_contrib = contrib[0]
             grad_flat = self.agg if self.is_master else None
         else:

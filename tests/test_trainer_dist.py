@@ -285,3 +285,67 @@ def test_bf16_wire_rejects_qsgd():
             num_classes=10, in_channels=1, device=torch.device("cpu"),
             wire_dtype="bf16",
         )
+
+
+def _run_rank_persistent(rank, world, port, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.set_num_threads(2)
+    import time
+
+    from atomo_amd.codings import make_codec
+    from atomo_amd.data import make_loaders
+    from atomo_amd.parallel import Comm, PSTrainer
+
+    comm = Comm(backend="gloo", device=torch.device("cpu"))
+    trainer = PSTrainer(
+        model_name="LeNet", codec=make_codec("svd", rank=3), comm=comm,
+        lr=0.05, momentum=0.9, num_classes=10, in_channels=1, seed=7,
+        device=torch.device("cpu"), comm_type="P2P", num_aggregate=2,
+    )
+    train, _ = make_loaders("mnist", 16, 16, torch.device("cpu"), seed=50 + rank)
+    it = iter(train)
+    if rank == 2:
+        # persistently slow COMPUTE (the self-skip path bypasses forward,
+        # so a skipping step is fast and the rank catches up)
+        trainer.model.register_forward_pre_hook(
+            lambda m, inp: time.sleep(0.25)
+        )
+    t0 = time.perf_counter()
+    for i in range(14):
+        x, y = next(it)
+        trainer.train_step(x, y)
+    total = time.perf_counter() - t0
+    q.put((rank, total, trainer.skip_count, getattr(comm, "stale_drops", 0)))
+    comm.barrier()
+    comm.close()
+
+
+def test_dist_persistent_straggler_self_skips():
+    """The tag-77 kill redesigned: a PERSISTENTLY slow rank must fall to
+    the self-skip path (shipping zero packets, skipping compute) so the
+    PS keeps running at the fast ranks' pace instead of throttling at
+    the pipeline depth."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    world, port = 3, _port(29645)
+    procs = [
+        ctx.Process(target=_run_rank_persistent, args=(r, world, port, q))
+        for r in range(world)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, total, skips, stale = q.get(timeout=240)
+        results[rank] = (total, skips, stale)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    # 14 steps with a 250 ms/step straggler: full-sync would cost >3.5 s;
+    # even a depth-6 pipeline WITHOUT self-skip throttles to ~2.2 s.
+    ps_total = results[0][0]
+    assert ps_total < 1.5, results
+    assert results[2][1] > 0  # the slow rank actually self-skipped
