@@ -67,12 +67,12 @@ SMALL_SM = 64      # LDS-Jacobi (64-variant) + batched_gram kernel
 # sm > that goes to batched hipSOLVER syevd on device (ResNet-50/152 have
 # 1x1-conv folds up to 2048x1024 — host LAPACK costs 30-60 ms each there).
 J128_SM = int(os.environ.get("ATOMO_JACOBI_CAP", "64"))
-# randomized mode routes folds ABOVE this to the randomized solver too:
-# the LDS Jacobi is one-WG-per-matrix and its launch runtime is the
-# single biggest matrix's round count (sm=64: ~0.5 ms on one CU), while
-# a 33..64 fold amortizes into the solver's batched GEMMs + merged
-# dense eigh for ~nothing (same unbiased-projection semantics)
-JAC_SM_RAND = int(os.environ.get("ATOMO_JACOBI_SM", "32"))
+# randomized mode routes folds ABOVE this to the randomized solver
+# (ATOMO_JACOBI_SM, read at encoder init).  A/B on one box: keeping the
+# sm=64 folds on the fork/join LDS Jacobi (default 64) beats routing
+# them to the randomized solver (32) by ~0.2 ms on ResNet-18 — the
+# Jacobi runs concurrently with the solver GEMMs, while an extra tiny
+# solver size-group adds serial launches.
 SOLVER_SM = 4096
 
 # measured per-matrix costs (ms) on MI355X + EPYC host, fp32 — used ONLY
@@ -255,7 +255,11 @@ class BatchedSVDEncoder:
             # group-level routing for big folds: batched hipSOLVER syevd is
             # nearly count-free per call, pooled host LAPACK wins for one or
             # two small-ish matrices (it overlaps the device Jacobi)
-            jac_cap = SMALL_SM if self.exact_eigh else JAC_SM_RAND
+            jac_cap = (
+                SMALL_SM
+                if self.exact_eigh
+                else int(os.environ.get("ATOMO_JACOBI_SM", "64"))
+            )
             self._jac_cap = jac_cap
             counts = defaultdict(int)
             for i, s in enumerate(specs):

@@ -179,13 +179,18 @@ def test_randomized_big_folds_match_topr(dev):
     steps, truncation mode must reproduce the top-r SVD reconstruction
     within the subspace tolerance."""
     torch.manual_seed(2)
-    # (128,64) has sm=64: above ATOMO_JACOBI_SM=32, so randomized mode
-    # routes it to the solver too (the LDS-Jacobi pole is the biggest
-    # matrix's round count)
-    shapes = [(128, 64, 1, 1), (256, 128, 1, 1), (512, 256, 1, 1),
-              (1024, 512, 1, 1)]
-    codec, specs, enc, flat, grads, wire = _build(dev, shapes, rank=3,
-                                                  exact_eigh=False)
+    # with ATOMO_JACOBI_SM=32 the (128,64) fold (sm=64) routes to the
+    # randomized solver too — exercises the 33..64 routing option
+    import os as _os
+
+    _os.environ["ATOMO_JACOBI_SM"] = "32"
+    try:
+        shapes = [(128, 64, 1, 1), (256, 128, 1, 1), (512, 256, 1, 1),
+                  (1024, 512, 1, 1)]
+        codec, specs, enc, flat, grads, wire = _build(dev, shapes, rank=3,
+                                                      exact_eigh=False)
+    finally:
+        _os.environ.pop("ATOMO_JACOBI_SM", None)
     assert len(enc.solver_layers) == 4, "33..64 folds join the solver path"
     assert not enc.exact_eigh and enc._rsvd_groups
     _decaying_grads(flat, grads, specs)
