@@ -253,8 +253,6 @@ class PSTrainer:
                     and len(enc.kernel_set) == len(self.wc.specs)
                     and self.codec.generator is None
                 )
-        # backward-hook overlap (reference *Split capability): per-layer
-        # encode work on a side stream while backward continues.
         # SPLIT graphs at world>1 (collective-gather mode, colocated):
         # capture (zero+fwd/bwd+encode) and, on the PS, (decode+apply) as
         # two graphs with the eager RCCL gather/broadcast between replays
@@ -285,9 +283,9 @@ class PSTrainer:
         self._graphB = None
         self._graphB_lr = None
         if self.use_graph and not self.graph_whole and not self.graph_split:
-            # capture ineligible (host layers / pinned RNG / qsgd):
-            # backward-hook overlap beats a fwd/bwd-only graph for the
-            # svd encode path, so let overlap win
+            # capture ineligible (host layers / pinned RNG / exact-eigh
+            # oracle / amp / Adam): backward-hook overlap beats a
+            # fwd/bwd-only graph for the svd encode path, so let it win
             self.use_graph = False
         self.overlap = (
             bool(overlap)
@@ -318,16 +316,7 @@ class PSTrainer:
                 if self.wc.reducible:
                     t.add_scalar("msg_bytes", 4.0 * self.wc.total_words)
                 self._last_contrib = self.num_workers
-                self.step_num += 1
-                if self.watchdog is not None:
-                    self.watchdog.step()
-                if self.step_num % self.shrink_freq == 0:
-                    self.lr *= self.lr_shrinkage
-                if (
-                    self.checkpoint_freq > 0
-                    and self.step_num % self.checkpoint_freq == 0
-                ):
-                    self.save_checkpoint()
+                self._post_step()
                 return self.last_loss
         if self.graph_split and self.step_num >= 1:
             # step 0 runs the standard eager path so every decode/apply
