@@ -313,7 +313,9 @@ class PSTrainer:
             with t.phase("comp"):
                 done = self._whole_step_graphed(x, y)
             if done:
-                if self.wc.reducible:
+                if self.wc.reducible or self.wc._qsgd_tables is not None:
+                    # fixed-layout wires ship every word each step (the
+                    # svd codec's variable packets are counted on device)
                     t.add_scalar("msg_bytes", 4.0 * self.wc.total_words)
                 self._last_contrib = self.num_workers
                 self._post_step()
@@ -599,6 +601,8 @@ class PSTrainer:
                     self.last_loss = float(loss.detach())
                 if not self.wc.reducible:
                     self.wc.encode_all(self.wire, flat_grad=self.flat_grad)
+        if self.wc.reducible or self.wc._qsgd_tables is not None:
+            t.add_scalar("msg_bytes", 4.0 * self.wc.total_words)
         with t.phase("comm"):
             if self.wc.reducible:
                 self.comm.reduce_sum(self.wire, dst=0)
