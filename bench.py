@@ -239,7 +239,14 @@ def main():
         print(
             json.dumps(
                 {
-                    "metric": "images/sec (ResNet-18 SVD-r=3 PS data-parallel)",
+                    "metric": (
+                        "images/sec (ResNet-18 SVD-r=3 PS data-parallel)"
+                        if (a.network, a.code, a.svd_rank)
+                        == ("ResNet18", "svd", 3)
+                        else f"images/sec ({a.network} {a.code}"
+                        + (f"-r{a.svd_rank}" if a.code == "svd" else "")
+                        + " PS data-parallel)"
+                    ),
                     "value": images_per_sec,
                     "unit": "images/s",
                     "n_gpus": n_gpus,
