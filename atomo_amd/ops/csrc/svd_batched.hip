@@ -1,10 +1,10 @@
 // Batched SVD-encode kernels for MI355X (gfx950).
 //
 // The SVD encode path (SURVEY §2.10 row "LA.svd") is restructured for the
-// GPU as: per-layer Gram matrices -> tiny host eigensolves -> selection
-// GEMMs writing the wire packets.  These kernels batch the two GPU phases
-// over ALL layers of a model in ONE launch each, driven by a descriptor
-// table, so a ~60-layer model costs 2 launches instead of ~250:
+// GPU as: per-layer Gram matrices -> batched eigensolves -> selection
+// writing the wire packets.  These kernels batch the two small-fold GPU
+// phases over ALL small layers of a model in ONE launch each, driven by a
+// descriptor table, so a ~60-layer model costs 2 launches instead of ~250:
 //
 //   batched_gram_kernel   G_l = A_l^T A_l (tall, n<=64) or A_l A_l^T
 //                         (wide, m<=64); LDS-staged row chunks, per-thread
@@ -14,8 +14,12 @@
 //                         step-invariant) written STRAIGHT into the wire
 //                         region; tile 0 also scatters header/s/small-factor.
 //
-// Layers with a big small-dim (> 64) or odd zero-padding stay on rocBLAS
-// GEMMs in python (a handful per model).
+// Layers with a big small-dim (> 64, the randomized-solver folds) run
+// their Grams and selection as shape-grouped rocBLAS bmms in python
+// (svd_encoder._compute_big_grams / _sel_big_folds); odd zero-padded
+// folds stay on per-layer rocBLAS GEMMs.  The staged facT/sel blocks use
+// a FIXED r_max stride, zero past the sampled r_hat, so the fixed-shape
+// GEMM consumers need no host sync.
 //
 // Descriptor layout (int64, per layer, GD_N words):
 //   [0] a_off      offset of the 2-D fold (m x n, row-major) in flat_grad
