@@ -398,7 +398,11 @@ __global__ void fused_sgd_kernel(float* __restrict__ p,
                                  float* __restrict__ buf, int64_t n, float lr,
                                  float momentum, float weight_decay,
                                  bool nesterov, float dampening,
-                                 float grad_scale) {
+                                 float grad_scale,
+                                 const float* __restrict__ lr_dev) {
+  // lr from device memory when provided: a captured pinned-memory copy
+  // refreshes it per hipGraph replay, so lr shrinkage needs no recapture
+  if (lr_dev != nullptr) lr = lr_dev[0];
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     float pi = p[i];
@@ -535,12 +539,12 @@ void atomo_svd_decode_batched_launch(const float* stacked, int64_t row_stride,
 void atomo_fused_sgd_launch(float* p, const float* g, float* buf, int64_t n,
                             float lr, float momentum, float weight_decay,
                             bool nesterov, float dampening, float grad_scale,
-                            hipStream_t stream) {
+                            const float* lr_dev, hipStream_t stream) {
   const int block = 256;
   const int grid = grid_for(n, block);
   hipLaunchKernelGGL(fused_sgd_kernel, dim3(grid), dim3(block), 0, stream, p,
                      g, buf, n, lr, momentum, weight_decay, nesterov,
-                     dampening, grad_scale);
+                     dampening, grad_scale, lr_dev);
 }
 
 }  // extern "C"

@@ -19,7 +19,8 @@ void atomo_qsgd_unpack_acc_launch(const float*, const uint32_t*, float*,
 void atomo_svd_decode_acc_launch(const float*, float*, int, int64_t, int, int,
                                  int, hipStream_t);
 void atomo_fused_sgd_launch(float*, const float*, float*, int64_t, float,
-                            float, float, bool, float, float, hipStream_t);
+                            float, float, bool, float, float, const float*,
+                            hipStream_t);
 void atomo_fused_adam_launch(float*, const float*, float*, float*, float*,
                              int64_t, float, float, float, float, float,
                              float, float, float, hipStream_t);
@@ -120,17 +121,21 @@ void svd_decode_acc(torch::Tensor regions, torch::Tensor out2d, int64_t m,
 
 void fused_sgd(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
                double momentum, double weight_decay, bool nesterov,
-               double dampening, double grad_scale) {
+               double dampening, double grad_scale, torch::Tensor lr_dev) {
   check_f32_cuda(p, "p");
   check_f32_cuda(g, "g");
   const int64_t n = p.numel();
   TORCH_CHECK(g.numel() == n, "grad size mismatch");
   if (momentum != 0.0)
     TORCH_CHECK(buf.numel() == n, "momentum buffer size mismatch");
+  if (lr_dev.numel())
+    TORCH_CHECK(lr_dev.is_cuda() && lr_dev.scalar_type() == torch::kFloat32,
+                "lr_dev must be cuda fp32");
   atomo_fused_sgd_launch(p.data_ptr<float>(), g.data_ptr<float>(),
                          buf.numel() ? buf.data_ptr<float>() : nullptr, n,
                          (float)lr, (float)momentum, (float)weight_decay,
                          nesterov, (float)dampening, (float)grad_scale,
+                         lr_dev.numel() ? lr_dev.data_ptr<float>() : nullptr,
                          cur_stream());
 }
 
@@ -395,7 +400,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("qsgd_unpack_acc", &qsgd_unpack_acc, "QSGD unpack+accumulate");
   m.def("svd_decode_acc", &svd_decode_acc,
         "fused rank-k SVD decode+accumulate over W packets");
-  m.def("fused_sgd", &fused_sgd, "fused flat SGD apply");
+  m.def("fused_sgd", &fused_sgd, py::arg("p"), py::arg("g"), py::arg("buf"),
+        py::arg("lr"), py::arg("momentum"), py::arg("weight_decay"),
+        py::arg("nesterov"), py::arg("dampening"), py::arg("grad_scale"),
+        py::arg("lr_dev") = torch::Tensor(), "fused flat SGD apply");
   m.def("fused_adam", &fused_adam, "fused flat Adam apply");
   m.def("rocsolver_eigh_batched", &rocsolver_eigh_batched,
         "rocSOLVER batched Jacobi eigensolver (fp32, in-place)");

@@ -28,7 +28,11 @@ def fused_sgd(
     nesterov: bool = False,
     dampening: float = 0.0,
     grad_scale: float = 1.0,
+    lr_dev: Optional[torch.Tensor] = None,
 ) -> None:
+    """``lr_dev`` (cuda fp32 scalar) overrides ``lr`` at kernel
+    execution time — under a hipGraph a captured pinned-memory copy into
+    it makes lr shrinkage replay-safe (no recapture)."""
     assert p.is_cuda
     ext().fused_sgd(
         p,
@@ -40,4 +44,5 @@ def fused_sgd(
         bool(nesterov),
         float(dampening),
         float(grad_scale),
+        lr_dev if lr_dev is not None else torch.empty(0, device=p.device),
     )

@@ -226,8 +226,14 @@ class PSTrainer:
         # device SVD path (device sampler seed rides pinned memory so each
         # replay draws fresh atoms).  Collectives are no-ops at world==1.
         self._wgraph = None
-        self._wgraph_lr = None
         self.graph_whole = False
+        # device-resident lr for the fused apply: a captured pinned copy
+        # re-reads it per replay, so lr shrinkage never forces a recapture
+        if self.device.type == "cuda" and self.is_master:
+            self._lr_host = torch.zeros(1, pin_memory=True)
+            self._lr_dev = torch.zeros(1, device=self.device)
+        else:
+            self._lr_host = self._lr_dev = None
         if self.use_graph and self.amp:
             # the captured bodies run fp32; don't silently drop autocast
             self.use_graph = False
@@ -281,7 +287,6 @@ class PSTrainer:
                 )
         self._graphA = None
         self._graphB = None
-        self._graphB_lr = None
         if self.use_graph and not self.graph_whole and not self.graph_split:
             # capture ineligible (host layers / pinned RNG / exact-eigh
             # oracle / amp / Adam): backward-hook overlap beats a
@@ -558,11 +563,13 @@ class PSTrainer:
 
     def _graphB_step(self) -> bool:
         """PS-side decode+apply graph (capture needs kernels warmed by an
-        eager step 0; recaptures when lr changes)."""
-        if self._graphB is None or self._graphB_lr != self.lr:
+        eager step 0; lr rides device memory, so shrinkage never forces a
+        recapture)."""
+        if self._graphB is None:
             g = None
             try:
                 self.opt.lr = self.lr
+                self._lr_host[0] = self.lr
                 g = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(g, capture_error_mode="thread_local"):
                     if self.wc.reducible:
@@ -572,7 +579,6 @@ class PSTrainer:
                         self.wc.decode_all(self.gather_buf, self.agg)
                         self._apply(self.agg)
                 self._graphB = g
-                self._graphB_lr = self.lr
             except Exception as exc:
                 print(
                     f"[atomo] split-graph B capture failed ({exc}); eager",
@@ -582,6 +588,7 @@ class PSTrainer:
                 self._graphB = None
                 self.graph_split = False
                 return False
+        self._lr_host[0] = self.lr  # replay re-reads the pinned lr
         self._graphB.replay()
         return True
 
@@ -640,14 +647,15 @@ class PSTrainer:
         self._apply(grad)
 
     def _whole_step_graphed(self, x: torch.Tensor, y: torch.Tensor) -> bool:
-        """Capture-once/replay whole-step graph.  Returns False (and
+        """Capture-once/replay whole-step graph (lr and sampler seeds ride
+        pinned->device copies, so replays track both).  Returns False (and
         disables itself) if capture fails — caller falls back to the
         standard path."""
-        if self._wgraph is None or self._wgraph_lr != self.lr:
+        if self._wgraph is None:
             g = None
             try:
-                self._wgraph = None
                 self.opt.lr = self.lr
+                self._lr_host[0] = self.lr
                 self._static_x = x.clone()
                 self._static_y = y.clone()
                 side = torch.cuda.Stream()
@@ -662,7 +670,6 @@ class PSTrainer:
                 with torch.cuda.graph(g, capture_error_mode="thread_local"):
                     self._step_body()
                 self._wgraph = g
-                self._wgraph_lr = self.lr
             except Exception as exc:
                 print(
                     f"[atomo] whole-step hipGraph capture failed ({exc}); "
@@ -675,6 +682,7 @@ class PSTrainer:
                 return False
         self._static_x.copy_(x)
         self._static_y.copy_(y)
+        self._lr_host[0] = self.lr  # replay re-reads the pinned lr
         self.wc.advance_seeds()  # captured H2D copy re-reads pinned seeds
         self._wgraph.replay()
         if self.defer_loss:
@@ -726,6 +734,11 @@ class PSTrainer:
         if self.flat.is_cuda and type(self.opt).__name__ == "ExternalSGD":
             from ..ops import optim_ops
 
+            lr_dev = None
+            if torch.cuda.is_current_stream_capturing():
+                # captured copy: replays re-read the pinned lr scalar
+                self._lr_dev.copy_(self._lr_host, non_blocking=True)
+                lr_dev = self._lr_dev
             optim_ops.fused_sgd(
                 self.flat,
                 grad_flat,
@@ -736,6 +749,7 @@ class PSTrainer:
                 nesterov=self.opt.nesterov,
                 dampening=self.opt.dampening,
                 grad_scale=scale,
+                lr_dev=lr_dev,
             )
         elif type(self.opt).__name__ == "ExternalAdam":
             self.opt.lr = self.lr

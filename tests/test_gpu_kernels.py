@@ -230,3 +230,52 @@ def test_qsgd_terngrad_batched_gpu(dev):
         uniq = vals.unique()
         for v in uniq:
             assert torch.isclose(v.abs(), mx, atol=1e-6) or v == 0
+
+
+def test_fused_sgd_device_lr(dev):
+    """lr_dev overrides the scalar lr at execution time (the hipGraph
+    replay path for lr shrinkage)."""
+    from atomo_amd.ops import optim_ops
+
+    torch.manual_seed(3)
+    p1 = torch.randn(1000, device=dev)
+    p2 = p1.clone()
+    g = torch.randn(1000, device=dev)
+    b1 = torch.zeros(1000, device=dev)
+    b2 = torch.zeros(1000, device=dev)
+    optim_ops.fused_sgd(p1, g, b1, lr=0.25, momentum=0.9)
+    lr_dev = torch.tensor([0.25], device=dev)
+    optim_ops.fused_sgd(p2, g, b2, lr=999.0, momentum=0.9, lr_dev=lr_dev)
+    assert torch.equal(p1, p2)  # scalar lr ignored when lr_dev given
+
+
+def test_gpu_graph_lr_shrinkage_no_recapture(dev):
+    """lr shrinkage under the whole-step graph: the SAME captured graph
+    keeps replaying (no recapture) and the shrunk lr actually reaches the
+    fused apply via device memory."""
+    from atomo_amd.codings import make_codec
+    from atomo_amd.data import make_loaders
+    from atomo_amd.parallel import Comm, PSTrainer
+
+    comm = Comm(device=dev)
+    trainer = PSTrainer(
+        model_name="ResNet18", codec=make_codec("svd", rank=3), comm=comm,
+        lr=0.04, momentum=0.9, lr_shrinkage=0.5, shrink_freq=3,
+        num_classes=10, in_channels=3, seed=11, device=dev, use_graph=True,
+    )
+    assert trainer.graph_whole
+    train, _ = make_loaders("cifar10", 64, 64, dev, seed=4)
+    it = iter(train)
+    for _ in range(4):
+        x, y = next(it)
+        trainer.train_step(x, y)
+    g_obj = trainer._wgraph
+    assert g_obj is not None
+    for _ in range(8):
+        x, y = next(it)
+        trainer.train_step(x, y)
+    assert trainer._wgraph is g_obj  # never recaptured
+    assert abs(trainer.lr - 0.04 * 0.5 ** 4) < 1e-9  # 12 steps / freq 3
+    assert torch.isfinite(trainer.flat).all()
+    # the device-side lr matches the shrunk host lr after the last replay
+    assert abs(float(trainer._lr_host[0]) - trainer.lr) < 1e-9
