@@ -74,7 +74,17 @@ class ExternalAdam:
     def load_state_dict(self, sd):
         for k in ("lr", "betas", "eps", "weight_decay", "amsgrad", "t"):
             setattr(self, k, sd[k])
-        self.exp_avg = sd["exp_avg"].to(self.p.device)
-        self.exp_avg_sq = sd["exp_avg_sq"].to(self.p.device)
+        # in place where shapes match (stable storage for any held refs)
+        for name in ("exp_avg", "exp_avg_sq"):
+            cur, new = getattr(self, name), sd[name].to(self.p.device)
+            if cur is not None and cur.shape == new.shape:
+                cur.copy_(new)
+            else:
+                setattr(self, name, new)
         if sd["max_exp_avg_sq"] is not None:
-            self.max_exp_avg_sq = sd["max_exp_avg_sq"].to(self.p.device)
+            new = sd["max_exp_avg_sq"].to(self.p.device)
+            cur = getattr(self, "max_exp_avg_sq", None)
+            if cur is not None and cur.shape == new.shape:
+                cur.copy_(new)
+            else:
+                self.max_exp_avg_sq = new

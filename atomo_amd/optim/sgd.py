@@ -74,4 +74,8 @@ class ExternalSGD:
         self.nesterov = sd["nesterov"]
         self.dampening = sd["dampening"]
         if sd["buf"] is not None:
-            self.buf = sd["buf"].to(self.p.device)
+            if self.buf is not None and self.buf.shape == sd["buf"].shape:
+                # in place: a captured hipGraph holds this buffer's pointer
+                self.buf.copy_(sd["buf"].to(self.p.device))
+            else:
+                self.buf = sd["buf"].to(self.p.device)

@@ -58,3 +58,21 @@ def test_sgd_state_roundtrip():
     mine.step(grads[1])
     other.step(grads[1])
     assert torch.allclose(flat, flat2)
+
+
+def test_sgd_state_load_is_in_place():
+    """A captured hipGraph holds the momentum buffer's pointer: loading
+    optimizer state must copy INTO the existing tensor."""
+    import torch
+
+    from atomo_amd.optim import make_optimizer
+
+    flat = torch.zeros(16)
+    opt = make_optimizer("sgd", flat, lr=0.1, momentum=0.9)
+    buf_before = opt.buf
+    opt.buf.fill_(1.0)
+    sd = opt.state_dict()
+    opt.buf.fill_(5.0)
+    opt.load_state_dict(sd)
+    assert opt.buf is buf_before
+    assert torch.all(opt.buf == 1.0)
