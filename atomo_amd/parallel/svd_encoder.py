@@ -561,10 +561,17 @@ class BatchedSVDEncoder:
                     self._rsvd_trace_dev = torch.zeros(
                         len(self.solver_layers), device=dev
                     )
+                    self._rsvd_warm = False
                     # precomputed index tensors: ONE gather / scatter kernel
                     # per group instead of O(layers) small slice copies (a
                     # 97-layer python loop costs milliseconds in launches)
                     for sm, idxs in sorted(by_sm.items()):
+                        # STABLE warm-basis buffer (written in place): a
+                        # captured graph reads/writes this exact storage,
+                        # so warm starts keep compounding across replays
+                        self._rsvd_Q[sm] = torch.zeros(
+                            len(idxs), sm, b, device=dev
+                        )
                         g0 = torch.arange(sm * sm, dtype=torch.int64)
                         gather = torch.cat(
                             [g0 + self.gram_offsets[i] for i in idxs]
@@ -869,8 +876,9 @@ class BatchedSVDEncoder:
             g = self.grams.index_select(0, gather).view(B, sm, sm)
             g = 0.5 * (g + g.transpose(1, 2))
             trs.append(g.diagonal(dim1=1, dim2=2).sum(dim=1))
-            q = self._rsvd_Q.get(sm)
-            if q is None or q.shape[0] != B:
+            if self._rsvd_warm:
+                q = self._rsvd_Q[sm]
+            else:
                 q = torch.randn(
                     B, sm, b, generator=self._rsvd_gen, device=self.device
                 )
@@ -904,7 +912,8 @@ class BatchedSVDEncoder:
             lam = lam_all[row0 : row0 + B]  # (B, b) descending
             evecs = torch.bmm(q, w_all[row0 : row0 + B])
             row0 += B
-            self._rsvd_Q[sm] = evecs  # warm subspace for the next step
+            self._rsvd_Q[sm].copy_(evecs)  # in-place: warm subspace for
+            # the next step (graph-replay-stable storage)
             # single-kernel writebacks (evals tail slots stay zero from
             # init -> atoms beyond the subspace are never sampled)
             self.grams.index_copy_(0, scatter, evecs.reshape(-1))
