@@ -844,7 +844,7 @@ class BatchedSVDEncoder:
         ext().jacobi_dense(s, ev, n, b, sweeps)
         return ev, s
 
-    def _merged_lowdin(self, ys):
+    def _merged_lowdin(self, ys, sweeps: int = 3):
         """Orthonormalize the columns of every (B_g, sm_g, b) panel:
         Q = Y S^{-1/2} (Lowdin / symmetric orthogonalization) with S's eigh
         merged across groups into ONE dense-Jacobi launch.  Column
@@ -855,7 +855,7 @@ class BatchedSVDEncoder:
         s = torch.cat(
             [torch.bmm(y.transpose(1, 2), y) for y in ys], dim=0
         ).contiguous()
-        lam, v = self._dense_eigh(s, sweeps=3)
+        lam, v = self._dense_eigh(s, sweeps=sweeps)
         inv = lam.clamp(min=1e-6).rsqrt()
         row0, out = 0, []
         for y in ys:
@@ -889,11 +889,12 @@ class BatchedSVDEncoder:
         # second Lowdin pass for near-machine orthogonality (the decoded
         # atom is A v v^T / p: unbiased for the projection onto span(V)
         # only when V is orthonormal)
+        lsweeps = 8 if cold else 3
         for _ in range(2 if cold else 1):
             qs = self._merged_lowdin(
-                [torch.bmm(g, q) for g, q in zip(gs, qs)]
+                [torch.bmm(g, q) for g, q in zip(gs, qs)], sweeps=lsweeps
             )
-        qs = self._merged_lowdin(qs)
+        qs = self._merged_lowdin(qs, sweeps=lsweeps)
         # Rayleigh-Ritz: T = Q^T G Q, merged eigh, Ritz pairs (lam, Q W)
         ts = torch.cat(
             [
@@ -902,7 +903,7 @@ class BatchedSVDEncoder:
             ],
             dim=0,
         ).contiguous()
-        lam_all, w_all = self._dense_eigh(ts, sweeps=5)
+        lam_all, w_all = self._dense_eigh(ts, sweeps=8 if cold else 5)
         lam_all = lam_all.clamp(min=0.0)
         row0 = 0
         for (sm, idxs, gather, scatter, evi, tails), q, tr in zip(
@@ -922,6 +923,7 @@ class BatchedSVDEncoder:
                 0, tails, (tr - lam.sum(dim=1)).clamp(min=0.0)
             )
             self._rsvd_trace_dev.index_copy_(0, tails, tr)
+        self._rsvd_warm = True
 
     def _solve_big_folds_exact(self) -> None:
         """Oracle path (ATOMO_EXACT_EIGH=1): batched hipSOLVER syevd per
