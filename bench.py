@@ -171,6 +171,30 @@ def main():
         trainer.train_step(x, y)
 
     sync()
+    if comm.world > 1 and a.wire_dtype == "fp32" and not trainer.p2p:
+        # weights flow from rank 0's broadcast, so after a broadcast every
+        # rank must hold BITWISE-identical parameters (rank 0 runs one
+        # apply ahead between steps, hence the explicit sync here); a
+        # mismatch means silent comm/graph corruption — surface it before
+        # the timed region
+        import sys
+
+        import torch.distributed as dist
+
+        comm.broadcast(trainer.flat, src=0)
+        sig = trainer.flat.view(torch.int32).to(torch.int64).sum()
+        sigs = [torch.zeros_like(sig) for _ in range(comm.world)]
+        dist.all_gather(sigs, sig)
+        if comm.rank == 0:
+            ok = all(bool(torch.equal(s, sigs[0])) for s in sigs)
+            print(
+                json.dumps(
+                    {"consistency": "ok" if ok else "MISMATCH",
+                     "sigs": [int(s) for s in sigs]}
+                ),
+                file=sys.stderr,
+                flush=True,
+            )
     trainer.timers.reset()
     trainer.wc.reset_device_msg_bytes()  # count timed steps only
     if a.phase_log:
